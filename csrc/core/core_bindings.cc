@@ -1,0 +1,183 @@
+// pybind11 bindings for the CPU DPF core -> python module gpudpf._core.
+// Keys cross the boundary as int32[524] numpy arrays (the torch wrapper in
+// gpudpf/dpf.py converts to/from torch tensors zero-copy).
+
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <atomic>
+#include <cstring>
+#include <thread>
+#include <vector>
+
+#include "dpf_core.h"
+
+namespace py = pybind11;
+using namespace gpudpf;
+
+namespace {
+
+using KeyArr = py::array_t<std::int32_t, py::array::c_style>;
+
+KeyArr key_to_array(const DpfKey& k) {
+  KeyArr arr(kKeyInts);
+  key_serialize(k, arr.mutable_data());
+  return arr;
+}
+
+DpfKey key_from_array(const KeyArr& arr) {
+  if (arr.size() != kKeyInts) throw std::invalid_argument("key must have 524 int32s");
+  DpfKey k;
+  key_deserialize(arr.data(), k);
+  return k;
+}
+
+py::tuple gen(u64 alpha, u64 n, py::bytes seed, int prf_method) {
+  std::string s = seed;
+  KeyRng rng(reinterpret_cast<const unsigned char*>(s.data()), s.size());
+  DpfKey k0, k1;
+  dpf_gen(alpha, /*beta=*/1, n, prf_method, rng, k0, k1);
+  return py::make_tuple(key_to_array(k0), key_to_array(k1));
+}
+
+// Full-domain expansion to low-32 shares, natural order -> int32[n].
+py::array_t<std::int32_t> expand(const KeyArr& key, int prf_method) {
+  DpfKey k = key_from_array(key);
+  py::array_t<std::int32_t> out((py::ssize_t)k.n);
+  {
+    py::gil_scoped_release nogil;
+    dpf_expand_full(k, prf_method, reinterpret_cast<u32*>(out.mutable_data()));
+  }
+  return out;
+}
+
+// Batched expansion with a thread pool (CPU serving path; the reference's
+// CPU baseline is OpenMP-parallel google-DPF expansion,
+// paper/kernel/cpu/dpf_google/benchmark.cu:117-120).
+py::array_t<std::int32_t> expand_batch(const std::vector<KeyArr>& keys,
+                                       int prf_method, int num_threads) {
+  const size_t b = keys.size();
+  if (b == 0) return py::array_t<std::int32_t>(0);
+  std::vector<DpfKey> ks(b);
+  for (size_t i = 0; i < b; ++i) ks[i] = key_from_array(keys[i]);
+  const u64 n = ks[0].n;
+  for (auto& k : ks)
+    if (k.n != n) throw std::invalid_argument("keys must share a domain size");
+  py::array_t<std::int32_t> out({(py::ssize_t)b, (py::ssize_t)n});
+  u32* optr = reinterpret_cast<u32*>(out.mutable_data());
+  {
+    py::gil_scoped_release nogil;
+    if (num_threads <= 1) {
+      for (size_t i = 0; i < b; ++i)
+        dpf_expand_full(ks[i], prf_method, optr + i * n);
+    } else {
+      std::vector<std::thread> pool;
+      std::atomic<size_t> next{0};
+      for (int t = 0; t < num_threads; ++t) {
+        pool.emplace_back([&] {
+          for (size_t i; (i = next.fetch_add(1)) < b;)
+            dpf_expand_full(ks[i], prf_method, optr + i * n);
+        });
+      }
+      for (auto& th : pool) th.join();
+    }
+  }
+  return out;
+}
+
+u64 eval_point_low(const KeyArr& key, u64 idx, int prf_method) {
+  DpfKey k = key_from_array(key);
+  return (u64)(u32)dpf_eval_point(k, idx, prf_method);
+}
+
+py::tuple eval_point128(const KeyArr& key, u64 idx, int prf_method) {
+  DpfKey k = key_from_array(key);
+  u128 v = dpf_eval_point(k, idx, prf_method);
+  return py::make_tuple((u64)v, (u64)(v >> 64));
+}
+
+py::array_t<std::int32_t> eval_fused_cpu(const KeyArr& key,
+                                         py::array_t<std::int32_t, py::array::c_style | py::array::forcecast> table,
+                                         int prf_method) {
+  DpfKey k = key_from_array(key);
+  if (table.ndim() != 2 || (u64)table.shape(0) != k.n)
+    throw std::invalid_argument("table must be [n, entry_words]");
+  int ew = (int)table.shape(1);
+  py::array_t<std::int32_t> out(ew);
+  {
+    py::gil_scoped_release nogil;
+    dpf_eval_fused_cpu(k, prf_method, reinterpret_cast<const u32*>(table.data()),
+                       ew, reinterpret_cast<u32*>(out.mutable_data()));
+  }
+  return out;
+}
+
+// natural -> permuted row map as an int64[n] array (used by eval_init to
+// reorder the table, and its inverse by the one-hot output path).
+py::array_t<std::int64_t> leaf_perm_table(u64 n, int zlog) {
+  py::array_t<std::int64_t> out((py::ssize_t)n);
+  auto* p = out.mutable_data();
+  {
+    py::gil_scoped_release nogil;
+    for (u64 i = 0; i < n; ++i) p[i] = (std::int64_t)leaf_perm(n, zlog, i);
+  }
+  return out;
+}
+
+py::tuple prf(int method, u64 seed_lo, u64 seed_hi, u64 pos) {
+  u128 s = ((u128)seed_hi << 64) | seed_lo;
+  u128 r = prf_eval(method, s, (u128)pos);
+  return py::make_tuple((u64)r, (u64)(r >> 64));
+}
+
+py::bytes aes_block(py::bytes key, py::bytes block) {
+  std::string k = key, b = block;
+  if (k.size() != 16 || b.size() != 16)
+    throw std::invalid_argument("key and block must be 16 bytes");
+  unsigned char out[16];
+  aes128_encrypt_block(reinterpret_cast<const unsigned char*>(k.data()),
+                       reinterpret_cast<const unsigned char*>(b.data()), out);
+  return py::bytes(reinterpret_cast<char*>(out), 16);
+}
+
+KeyArr shard_subkey(const KeyArr& key, int prf_method, u64 rank, u64 world) {
+  DpfKey k = key_from_array(key);
+  DpfKey sub;
+  dpf_shard_subkey(k, prf_method, rank, world, sub);
+  return key_to_array(sub);
+}
+
+// AES GPU tables (5 x 256 u32: te0..te3, sbox) for upload to the device.
+py::array_t<std::int32_t> aes_gpu_tables() {
+  py::array_t<std::int32_t> out(5 * 256);
+  u32* p = reinterpret_cast<u32*>(out.mutable_data());
+  aes128_tables(p, p + 256, p + 512, p + 768, p + 1024);
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "gpudpf CPU core (keygen, reference eval, layout)";
+  m.def("gen", &gen, py::arg("alpha"), py::arg("n"), py::arg("seed"),
+        py::arg("prf_method"));
+  m.def("expand", &expand, py::arg("key"), py::arg("prf_method"));
+  m.def("expand_batch", &expand_batch, py::arg("keys"), py::arg("prf_method"),
+        py::arg("num_threads") = 1);
+  m.def("eval_point_low", &eval_point_low);
+  m.def("eval_point128", &eval_point128);
+  m.def("eval_fused_cpu", &eval_fused_cpu);
+  m.def("leaf_perm_table", &leaf_perm_table);
+  m.def("zlog_for_depth", &zlog_for_depth);
+  m.def("prf", &prf);
+  m.def("aes_block", &aes_block);
+  m.def("shard_subkey", &shard_subkey);
+  m.def("aes_gpu_tables", &aes_gpu_tables);
+  m.attr("KEY_INTS") = py::int_(kKeyInts);
+  m.attr("ENTRY_WORDS") = py::int_(kEntryWords);
+  m.attr("PRF_DUMMY") = py::int_((int)PRF_DUMMY);
+  m.attr("PRF_SALSA20") = py::int_((int)PRF_SALSA20);
+  m.attr("PRF_CHACHA20") = py::int_((int)PRF_CHACHA20);
+  m.attr("PRF_AES128") = py::int_((int)PRF_AES128);
+}

@@ -1,0 +1,301 @@
+// DPF key generation / evaluation / serialization / layout (see dpf_core.h).
+
+#include "dpf_core.h"
+
+#include <vector>
+
+namespace gpudpf {
+
+// ---------------------------------------------------------------------------
+// RNG
+// ---------------------------------------------------------------------------
+KeyRng::KeyRng(const unsigned char* seed, size_t len) {
+  std::seed_seq seq(seed, seed + len);
+  gen_.seed(seq);
+}
+
+u128 KeyRng::next_u128() {
+  return ((u128)gen_() << 64) | (u128)gen_();
+}
+
+u128 KeyRng::next_odd_u128() { return next_u128() | 1; }
+
+// ---------------------------------------------------------------------------
+// Key generation (iterative; O(log n) PRF calls)
+// ---------------------------------------------------------------------------
+// Eval recurrence (per key k, index idx):
+//   key = root
+//   for i = depth-1 .. 0:  b = idx & 1
+//     key = PRF(key, b) + cw[key & 1][i*2 + b]   (mod 2^128)
+//     idx >>= 1
+// Level i consumes index bit (depth-1-i).  The generator walks the on-path
+// seeds (s0, s1) top-down, maintaining the invariant that after the level
+// using payload beta_i we have s0 - s1 == beta_i (odd => parities differ,
+// so the two servers always select opposite correction words on the path;
+// off the path the seeds coincide and all contributions cancel).
+void dpf_gen(u64 alpha, u128 beta, u64 n, int prf_method, KeyRng& rng,
+             DpfKey& k0, DpfKey& k1) {
+  if (n < 2 || (n & (n - 1)) != 0)
+    throw std::invalid_argument("n must be a power of two >= 2");
+  if (alpha >= n) throw std::invalid_argument("alpha must be < n");
+  const int depth = ilog2_u64(n);
+  if (depth > kMaxDepth) throw std::invalid_argument("n too large");
+
+  k0 = DpfKey{};
+  k1 = DpfKey{};
+  k0.depth = k1.depth = depth;
+  k0.n = k1.n = n;
+
+  // Per-level payloads: the innermost level (i = depth-1, consumed first)
+  // through i = 1 carry fresh random odd payloads; the outermost level
+  // (i = 0, consumed last) carries the caller's beta.
+  std::vector<u128> betas((size_t)depth);
+  betas[0] = beta;
+  for (int i = 1; i < depth; ++i) betas[(size_t)i] = rng.next_odd_u128();
+
+  // Root seeds with forced opposite parity (server 0 even, server 1 odd).
+  u128 s0 = rng.next_u128() & ~(u128)1;
+  u128 s1 = rng.next_u128() | 1;
+  k0.root = s0;
+  k1.root = s1;
+
+  for (int i = depth - 1; i >= 0; --i) {
+    const int level_bit = (int)((alpha >> (depth - 1 - i)) & 1);
+    const u128 beta_i = betas[(size_t)i];
+
+    // Correction word pair for this level.  Server with even seed adds
+    // cw[0], odd seed adds cw[1]; on the path parities differ, so the pair
+    // difference must cancel the PRF divergence and inject beta_i at the
+    // target bit.  Writing e = the even-seed server's value, o = the odd
+    // one's:  e_val - o_val == [bit == level_bit] * sgn * beta_i where sgn
+    // accounts for which server holds the even seed.
+    u128 cw_path[2];  // the codeword pair at b == level_bit (for the advance)
+    for (int b = 0; b < 2; ++b) {
+      u128 p0 = prf_eval(prf_method, s0, (u128)b);
+      u128 p1 = prf_eval(prf_method, s1, (u128)b);
+      u128 rnd = rng.next_u128();
+      // Let cw[sel(s0)] = rnd; solve cw[sel(s1)] so that
+      //   (p0 + cw[sel0]) - (p1 + cw[sel1]) == (b == level_bit) ? beta_i : 0
+      u128 target = (b == level_bit) ? beta_i : (u128)0;
+      u128 other = p0 + rnd - p1 - target;  // cw[sel1]
+      int sel0 = (int)(s0 & 1);
+      u128 cw_pair[2];
+      cw_pair[sel0] = rnd;
+      cw_pair[sel0 ^ 1] = other;
+      k0.cw[0][i * 2 + b] = k1.cw[0][i * 2 + b] = cw_pair[0];
+      k0.cw[1][i * 2 + b] = k1.cw[1][i * 2 + b] = cw_pair[1];
+      if (b == level_bit) {
+        cw_path[0] = cw_pair[0];
+        cw_path[1] = cw_pair[1];
+      }
+    }
+
+    // Advance the on-path seeds through this level.
+    u128 n0 = prf_eval(prf_method, s0, (u128)level_bit) +
+              cw_path[(size_t)(s0 & 1)];
+    u128 n1 = prf_eval(prf_method, s1, (u128)level_bit) +
+              cw_path[(size_t)(s1 & 1)];
+    s0 = n0;
+    s1 = n1;
+    if ((u128)(s0 - s1) != beta_i)
+      throw std::logic_error("dpf_gen: level invariant violated");
+    if (i > 0 && ((s0 ^ s1) & 1) == 0)
+      throw std::logic_error("dpf_gen: parity invariant violated");
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Evaluation
+// ---------------------------------------------------------------------------
+u128 dpf_eval_point(const DpfKey& k, u64 idx, int prf_method) {
+  u128 key = k.root;
+  u64 rem = idx;
+  for (int i = k.depth - 1; i >= 0; --i) {
+    int b = (int)(rem & 1);
+    u128 v = prf_eval(prf_method, key, (u128)b);
+    key = v + k.cw[(size_t)(key & 1)][i * 2 + b];
+    rem >>= 1;
+  }
+  return key;
+}
+
+namespace {
+
+// Expand both children of `seed` at eval-level i.  For AES, share one key
+// schedule across the two encryptions (the reference re-expands per call;
+// see SURVEY.md "hard parts" item 3).
+inline void expand_children(const DpfKey& k, int prf_method, u128 seed, int i,
+                            u128& c0, u128& c1) {
+  u128 p0, p1;
+  if (prf_method == PRF_AES128) {
+    unsigned char key[16], in0[16] = {0}, in1[16] = {0}, out[16];
+    u32 rk[44];
+    std::memcpy(key, &seed, 16);
+    aes128_expand_key(key, rk);
+    in1[0] = 1;
+    aes128_encrypt_block_rk(rk, in0, out);
+    std::memcpy(&p0, out, 16);
+    aes128_encrypt_block_rk(rk, in1, out);
+    std::memcpy(&p1, out, 16);
+  } else {
+    p0 = prf_eval(prf_method, seed, 0);
+    p1 = prf_eval(prf_method, seed, 1);
+  }
+  int sel = (int)(seed & 1);
+  c0 = p0 + k.cw[sel][i * 2 + 0];
+  c1 = p1 + k.cw[sel][i * 2 + 1];
+}
+
+}  // namespace
+
+// Full-domain expansion, natural order, O(n) PRF pairs (the reference's CPU
+// expansion is O(n log n) single calls: dpf_wrapper.cu:70-84).  Node arrays
+// are indexed by the partial natural index (consumed bit ell at weight
+// 2^ell), so leaves land in natural order directly.
+void dpf_expand_full(const DpfKey& k, int prf_method, u32* out) {
+  const int depth = k.depth;
+  std::vector<u128> cur(1, k.root), next;
+  for (int l = 0; l < depth; ++l) {
+    const int i = depth - 1 - l;  // eval-order level index
+    const u64 width = (u64)1 << l;
+    if (i == 0) {
+      // Leaf level: emit low-32 shares straight into out.
+      for (u64 v = 0; v < width; ++v) {
+        u128 c0, c1;
+        expand_children(k, prf_method, cur[v], i, c0, c1);
+        out[v] = (u32)c0;
+        out[v | ((u64)1 << l)] = (u32)c1;
+      }
+    } else {
+      next.resize(width * 2);
+      for (u64 v = 0; v < width; ++v) {
+        u128 c0, c1;
+        expand_children(k, prf_method, cur[v], i, c0, c1);
+        next[v] = c0;
+        next[v | ((u64)1 << l)] = c1;
+      }
+      cur.swap(next);
+    }
+  }
+  if (depth == 0) out[0] = (u32)k.root;
+}
+
+void dpf_eval_fused_cpu(const DpfKey& k, int prf_method, const u32* table,
+                        int entry_words, u32* out) {
+  std::vector<u32> shares(k.n);
+  dpf_expand_full(k, prf_method, shares.data());
+  for (int m = 0; m < entry_words; ++m) out[m] = 0;
+  for (u64 idx = 0; idx < k.n; ++idx) {
+    u32 s = shares[idx];
+    const u32* row = table + idx * (u64)entry_words;
+    for (int m = 0; m < entry_words; ++m) out[m] += s * row[m];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Serialization (wire format parity with the reference; dpf_wrapper.cu:26-46)
+// ---------------------------------------------------------------------------
+void key_serialize(const DpfKey& k, std::int32_t out[kKeyInts]) {
+  std::memset(out, 0, sizeof(std::int32_t) * kKeyInts);
+  u128* slots = reinterpret_cast<u128*>(out);
+  slots[0] = (u128)(unsigned)k.depth;
+  for (int i = 0; i < 64; ++i) {
+    slots[1 + i] = k.cw[0][i];
+    slots[65 + i] = k.cw[1][i];
+  }
+  slots[129] = k.root;
+  slots[130] = (u128)k.n;
+}
+
+void key_deserialize(const std::int32_t in[kKeyInts], DpfKey& k) {
+  const u128* slots = reinterpret_cast<const u128*>(in);
+  k.depth = (int)(u64)slots[0];
+  if (k.depth < 1 || k.depth > kMaxDepth)
+    throw std::invalid_argument("corrupt key: bad depth");
+  for (int i = 0; i < 64; ++i) {
+    k.cw[0][i] = slots[1 + i];
+    k.cw[1][i] = slots[65 + i];
+  }
+  k.root = slots[129];
+  k.n = (u64)slots[130];
+  if (k.n != ((u64)1 << k.depth))
+    throw std::invalid_argument("corrupt key: n != 2^depth");
+}
+
+// ---------------------------------------------------------------------------
+// Layout contract
+// ---------------------------------------------------------------------------
+int zlog_for_depth(int depth) {
+  // Z = workgroup size of the fused kernel (256 threads = 4 wave64), capped
+  // so each thread keeps a subtree of at least one leaf pair (DS >= 1).
+  int zlog = 8;
+  if (depth - 1 < zlog) zlog = depth - 1;
+  if (zlog < 0) zlog = 0;
+  return zlog;
+}
+
+namespace {
+inline u64 bitrev(u64 v, int bits) {
+  u64 r = 0;
+  for (int i = 0; i < bits; ++i) r |= ((v >> i) & 1) << (bits - 1 - i);
+  return r;
+}
+}  // namespace
+
+u64 leaf_perm(u64 n, int zlog, u64 idx) {
+  const int depth = ilog2_u64(n);
+  const int ds = depth - zlog;  // subtree splits per thread (>= 1)
+  const u64 t = bitrev(idx & ((n > 1 ? ((u64)1 << zlog) : 1) - 1), zlog);
+  const u64 mid = (ds >= 2) ? ((idx >> zlog) & (((u64)1 << (ds - 1)) - 1)) : 0;
+  const u64 j = bitrev(mid, ds - 1);
+  const u64 b = (idx >> (depth - 1)) & 1;
+  return (j << (zlog + 1)) | (t << 1) | b;
+}
+
+u64 leaf_perm_inv(u64 n, int zlog, u64 row) {
+  const int depth = ilog2_u64(n);
+  const int ds = depth - zlog;
+  const u64 b = row & 1;
+  const u64 t = (row >> 1) & (((u64)1 << zlog) - 1);
+  const u64 j = row >> (zlog + 1);
+  u64 idx = bitrev(t, zlog);
+  idx |= bitrev(j, ds - 1) << zlog;
+  idx |= b << (depth - 1);
+  return idx;
+}
+
+// ---------------------------------------------------------------------------
+// Sharding
+// ---------------------------------------------------------------------------
+void dpf_shard_subkey(const DpfKey& k, int prf_method, u64 rank, u64 world,
+                      DpfKey& out) {
+  if (world < 1 || (world & (world - 1)) != 0)
+    throw std::invalid_argument("world size must be a power of two");
+  if (rank >= world) throw std::invalid_argument("rank must be < world");
+  const int wlog = ilog2_u64(world);
+  if (wlog >= k.depth)
+    throw std::invalid_argument("world size too large for key depth");
+
+  out = DpfKey{};
+  out.depth = k.depth - wlog;
+  out.n = k.n >> wlog;
+  // Walk wlog levels from the root consuming rank bits LSB-first; the
+  // remaining correction words (eval levels 0..depth-wlog-1) carry over
+  // unchanged because eval level indices count from the leaf end.
+  u128 key = k.root;
+  for (int l = 0; l < wlog; ++l) {
+    const int i = k.depth - 1 - l;
+    const int b = (int)((rank >> l) & 1);
+    u128 v = prf_eval(prf_method, key, (u128)b);
+    key = v + k.cw[(size_t)(key & 1)][i * 2 + b];
+  }
+  out.root = key;
+  for (int i = 0; i < out.depth; ++i) {
+    for (int b = 0; b < 2; ++b) {
+      out.cw[0][i * 2 + b] = k.cw[0][i * 2 + b];
+      out.cw[1][i * 2 + b] = k.cw[1][i * 2 + b];
+    }
+  }
+}
+
+}  // namespace gpudpf

@@ -1,0 +1,28 @@
+// Host-side API of the gpudpf HIP kernel library (implemented in
+// dpf_kernels.hip, bound to python in hip_bindings.cc).
+#pragma once
+#include <cstdint>
+
+namespace gpudpf_hip {
+
+// Fused DPF expansion + table inner product (the production PIR path).
+//   keys:  device ptr, int32 [batch][524]   (wire-format keys)
+//   table: device ptr, u32 [n][16]          (leaf_perm-reordered rows)
+//   out:   device ptr, u32 [batch][16]
+//   aes_tabs: device ptr to 5*256 u32 AES tables (only read for AES128)
+void launch_fused(std::uintptr_t keys, std::uintptr_t table, std::uintptr_t out,
+                  std::uintptr_t aes_tabs, int batch, long long n, int depth,
+                  int zlog, int prf, std::uintptr_t stream);
+
+// Full expansion to low-32 one-hot shares, permuted row order
+//   out: device ptr, u32 [batch][n]  (row r = leaf_perm(idx))
+void launch_expand(std::uintptr_t keys, std::uintptr_t out,
+                   std::uintptr_t aes_tabs, int batch, long long n, int depth,
+                   int zlog, int prf, std::uintptr_t stream);
+
+// Naive per-leaf oracle (O(n log n) PRFs), natural order output.  Test-only.
+void launch_naive(std::uintptr_t keys, std::uintptr_t out,
+                  std::uintptr_t aes_tabs, int batch, long long n, int depth,
+                  int prf, std::uintptr_t stream);
+
+}  // namespace gpudpf_hip

@@ -1,0 +1,20 @@
+"""gpudpf — MI355X-native Distributed Point Function / 2-server PIR engine.
+
+A from-scratch AMD CDNA4 framework with the capabilities of
+facebookresearch/GPU-DPF: CPU key generation, hand-written HIP (gfx950)
+batched DPF evaluation with a fused table inner product, multi-GPU row
+sharding over RCCL/xGMI, and the batch-PIR co-design toolkit.
+
+Public API parity (reference dpf.py:35-137):
+
+    from gpudpf import DPF
+    d = DPF(prf=DPF.PRF_AES128)
+    k1, k2 = d.gen(k, n)
+    d.eval_init(table)          # [n, e<=16] int32, n a power of two >= 128
+    shares = d.eval_gpu([k1])   # [1, e] int32 secret shares
+"""
+
+from gpudpf.dpf import DPF  # noqa: F401
+from gpudpf.dist import ShardedDPF  # noqa: F401
+
+__version__ = "0.1.0"

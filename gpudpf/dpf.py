@@ -1,0 +1,217 @@
+"""User-facing DPF API (parity with the reference's dpf.DPF, dpf.py:35-137,
+plus capabilities the reference leaves as TODOs: arbitrary batch sizes
+without padding, GPU one-hot share output, runtime PRF choice per call
+path, >BATCH_SIZE batches in one launch)."""
+
+import os
+
+import numpy as np
+import torch
+
+from gpudpf import _core
+
+try:
+    from gpudpf import _hip
+
+    _HAS_HIP = True
+except ImportError:  # pragma: no cover - HIP runtime missing entirely
+    _hip = None
+    _HAS_HIP = False
+
+
+class DPF(object):
+    PRF_DUMMY = _core.PRF_DUMMY
+    PRF_SALSA20 = _core.PRF_SALSA20
+    PRF_CHACHA20 = _core.PRF_CHACHA20
+    PRF_AES128 = _core.PRF_AES128
+
+    ENTRY_SIZE = _core.ENTRY_WORDS  # 16 x u32 per table entry (padded)
+    BATCH_SIZE = 512                # reference-compatible chunking constant
+    MAX_LAUNCH_BATCH = 4096         # keys per kernel launch
+
+    DEFAULT_PRF = _core.PRF_AES128
+    KEY_INTS = _core.KEY_INTS       # 524 int32 = 2096 bytes per key
+
+    _PRF_NAMES = {
+        _core.PRF_DUMMY: "DUMMY",
+        _core.PRF_SALSA20: "SALSA20",
+        _core.PRF_CHACHA20: "CHACHA20",
+        _core.PRF_AES128: "AES128",
+    }
+
+    def __init__(self, prf=None, device=None):
+        self.prf_method = self.DEFAULT_PRF if prf is None else prf
+        self.prf_method_string = self._PRF_NAMES[self.prf_method]
+        self.device = device  # resolved lazily at eval_init
+
+        self.table = None               # natural-order table (as given)
+        self.table_num_entries = None
+        self.table_effective_entry_size = None
+        self._table_gpu = None          # leaf_perm-reordered [n,16] int32 on GPU
+        self._perm_gpu = None           # natural->row map (for one-hot unperm)
+        self._zlog = None
+        self._depth = None
+        self._aes_ptr = 0
+
+    # ------------------------------------------------------------------
+    # Client side
+    # ------------------------------------------------------------------
+    def gen(self, k, n):
+        """Generate the two server keys selecting index k of an n-entry
+        table.  Returns two int32[524] torch tensors (2096-byte keys)."""
+        if n & (n - 1) != 0:
+            raise Exception("Table num entries (%d) must be a power of two" % n)
+        if k >= n:
+            raise Exception(
+                "k (%d), the selected element, must be less than n (%d), the "
+                "number of entries in the table" % (k, n)
+            )
+        seed = os.urandom(128)
+        k1, k2 = _core.gen(k, n, seed, self.prf_method)
+        return [torch.from_numpy(k1), torch.from_numpy(k2)]
+
+    # ------------------------------------------------------------------
+    # Server side
+    # ------------------------------------------------------------------
+    def eval_init(self, table):
+        """Upload an [n, e] int32 table (n power-of-two >= 128, e <= 16).
+        Rows are reordered by the kernel layout contract (leaf_perm) and
+        padded to 16 words."""
+        self.table = table
+        self.table_num_entries = int(table.shape[0])
+        self.table_effective_entry_size = int(table.shape[1])
+        n, e = self.table_num_entries, self.table_effective_entry_size
+
+        if n < 128:
+            raise Exception("Table (%d) must have at least 128 elements" % n)
+        if n & (n - 1) != 0:
+            raise Exception("Table num entries (%d) must be a power of two" % n)
+        if e > self.ENTRY_SIZE:
+            raise Exception(
+                "Table entry dimension (%d) must be < %d" % (e, self.ENTRY_SIZE)
+            )
+
+        self._depth = n.bit_length() - 1
+        self._zlog = _core.zlog_for_depth(self._depth)
+
+        if self.device is None:
+            self.device = "cuda:0" if torch.cuda.is_available() else "cpu"
+        dev = torch.device(self.device)
+
+        padded = torch.zeros((n, self.ENTRY_SIZE), dtype=torch.int32)
+        padded[:, :e] = table.to(torch.int32)
+        perm = torch.from_numpy(_core.leaf_perm_table(n, self._zlog))
+        reordered = torch.empty_like(padded)
+        reordered[perm] = padded  # row perm[i] <- natural row i
+
+        if dev.type == "cuda":
+            self._table_gpu = reordered.to(dev).contiguous()
+            self._perm_gpu = perm.to(dev)
+            if self.prf_method == self.PRF_AES128:
+                self._aes_ptr = _hip.ensure_aes_tables(dev.index or 0)
+        else:
+            self._table_gpu = None
+            self._perm_gpu = None
+
+    def eval_free(self):
+        self._table_gpu = None
+        self._perm_gpu = None
+
+    def _keys_tensor(self, keys):
+        if isinstance(keys, torch.Tensor):
+            kt = keys
+            if kt.dim() == 1:
+                kt = kt.unsqueeze(0)
+        else:
+            kt = torch.stack([k.reshape(-1) for k in keys])
+        if kt.dtype != torch.int32 or kt.shape[1] != self.KEY_INTS:
+            raise Exception("keys must be int32[524] tensors")
+        n = int(kt[0, 520].item())  # u128 slot 130 low word
+        depth = int(kt[0, 0].item())
+        return kt.contiguous(), n, depth
+
+    def eval_gpu(self, keys, one_hot_only=False):
+        """Evaluate a batch of keys against the initialized table on the
+        GPU.  Returns [batch, e] int32 secret shares (CPU tensor), or the
+        raw [batch, n] one-hot shares if one_hot_only (a capability the
+        reference lists as a TODO, dpf.py:30)."""
+        if self._table_gpu is None:
+            raise Exception("Must call `eval_init` before `eval_gpu`")
+        if not _HAS_HIP:
+            raise Exception("gpudpf._hip extension is not available")
+        kt, n, depth = self._keys_tensor(keys)
+        if n != self.table_num_entries:
+            raise Exception(
+                "key domain (%d) does not match table (%d)" % (n, self.table_num_entries)
+            )
+        batch = kt.shape[0]
+        dev = self._table_gpu.device
+        stream = torch.cuda.current_stream(dev).cuda_stream
+        keys_gpu = kt.to(dev, non_blocking=True)
+
+        results = []
+        for lo in range(0, batch, self.MAX_LAUNCH_BATCH):
+            hi = min(batch, lo + self.MAX_LAUNCH_BATCH)
+            chunk = keys_gpu[lo:hi].contiguous()
+            b = hi - lo
+            if one_hot_only:
+                out = torch.empty((b, n), dtype=torch.int32, device=dev)
+                _hip.eval_expand(
+                    chunk.data_ptr(), out.data_ptr(), self._aes_ptr, b, n,
+                    depth, self._zlog, self.prf_method, stream,
+                )
+                # rows are in leaf_perm order; gather back to natural order
+                out = out.index_select(1, self._perm_gpu)
+            else:
+                out = torch.empty((b, self.ENTRY_SIZE), dtype=torch.int32, device=dev)
+                _hip.eval_fused(
+                    chunk.data_ptr(), self._table_gpu.data_ptr(), out.data_ptr(),
+                    self._aes_ptr, b, n, depth, self._zlog, self.prf_method,
+                    stream,
+                )
+                out = out[:, : self.table_effective_entry_size]
+            results.append(out)
+        res = torch.cat(results) if len(results) > 1 else results[0]
+        return res.cpu()
+
+    def eval_gpu_into(self, keys_gpu, out_gpu):
+        """Zero-copy serving path: keys already on device as [b,524] int32,
+        fused result written into out_gpu [b,16] int32 asynchronously on the
+        current stream (no host sync)."""
+        b = keys_gpu.shape[0]
+        dev = self._table_gpu.device
+        stream = torch.cuda.current_stream(dev).cuda_stream
+        _hip.eval_fused(
+            keys_gpu.data_ptr(), self._table_gpu.data_ptr(), out_gpu.data_ptr(),
+            self._aes_ptr, b, self.table_num_entries, self._depth, self._zlog,
+            self.prf_method, stream,
+        )
+
+    def eval_cpu(self, keys, one_hot_only=False, num_threads=None):
+        """CPU reference evaluation (O(n) PRF pairs per key; the reference's
+        CPU path is O(n log n), dpf_wrapper.cu:70-84)."""
+        kt, n, _depth = self._keys_tensor(keys)
+        if num_threads is None:
+            num_threads = min(32, os.cpu_count() or 1)
+        arrs = [kt[i].numpy() for i in range(kt.shape[0])]
+        one_hots = torch.from_numpy(
+            np.asarray(_core.expand_batch(arrs, self.prf_method, num_threads))
+        )
+        if one_hot_only:
+            return one_hots
+        if self.table is None:
+            raise Exception(
+                "Must call `eval_init` before `eval_cpu` with one_hot_only=False"
+            )
+        return torch.matmul(one_hots.to(torch.int64), self.table.to(torch.int64)).to(
+            torch.int32
+        )
+
+    def __repr__(self):
+        if self.table is None:
+            return "DPF(_uninitialized_, prf_method=%s)" % self.prf_method_string
+        return "DPF(entries=%d, entry_size=%d, prf_method=%s)" % (
+            self.table_num_entries,
+            self.table_effective_entry_size,
+            self.prf_method_string,
+        )

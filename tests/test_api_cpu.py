@@ -1,0 +1,83 @@
+"""DPF python API tests, CPU paths (mirrors the reference's python
+integration tests dpf.py:139-204 as pytest)."""
+
+import random
+
+import numpy as np
+import pytest
+import torch
+
+from gpudpf import DPF
+
+
+def test_cpu_dpf_one_hot():
+    N = 1024
+    dpf = DPF()
+    K = 42
+    k1, k2 = dpf.gen(K, N)
+    v1 = dpf.eval_cpu([k1], one_hot_only=True)
+    v2 = dpf.eval_cpu([k2], one_hot_only=True)
+    rec = (v1 - v2).numpy()
+    gt = np.zeros(rec.shape)
+    gt[:, K] = 1
+    assert np.linalg.norm(rec - gt) <= 1e-8
+
+
+def test_cpu_dpf_with_table():
+    N = 1024
+    dpf = DPF(device="cpu")
+    k1s, k2s, gt_indices = [], [], []
+    for _ in range(16):
+        indx = random.randint(0, N - 1)
+        gt_indices.append(indx)
+        k1, k2 = dpf.gen(indx, N)
+        k1s.append(k1)
+        k2s.append(k2)
+    table = torch.arange(N * 16, dtype=torch.int32).reshape(N, 16)
+    dpf.eval_init(table)
+    a = dpf.eval_cpu(k1s)
+    b = dpf.eval_cpu(k2s)
+    rec = (a - b).numpy()
+    gt = table[gt_indices, :].numpy()
+    assert np.linalg.norm(rec - gt) <= 1e-8
+
+
+def test_gen_validation():
+    dpf = DPF()
+    with pytest.raises(Exception):
+        dpf.gen(0, 1000)  # not a power of two
+    with pytest.raises(Exception):
+        dpf.gen(1024, 1024)  # k >= n
+
+
+def test_eval_init_validation():
+    dpf = DPF(device="cpu")
+    with pytest.raises(Exception):
+        dpf.eval_init(torch.zeros((64, 16), dtype=torch.int32))  # < 128 rows
+    with pytest.raises(Exception):
+        dpf.eval_init(torch.zeros((100, 16), dtype=torch.int32))  # not pow2
+    with pytest.raises(Exception):
+        dpf.eval_init(torch.zeros((128, 17), dtype=torch.int32))  # entry too big
+
+
+def test_key_size_constant():
+    dpf = DPF()
+    for n in [128, 8192, 1 << 20]:
+        k1, _ = dpf.gen(1, n)
+        assert int(np.prod(k1.shape)) * 4 == 2096
+
+
+def test_prf_choice():
+    for prf in [DPF.PRF_DUMMY, DPF.PRF_SALSA20, DPF.PRF_CHACHA20, DPF.PRF_AES128]:
+        dpf = DPF(prf=prf)
+        k1, k2 = dpf.gen(3, 128)
+        v = (dpf.eval_cpu([k1], one_hot_only=True) -
+             dpf.eval_cpu([k2], one_hot_only=True))
+        assert v[0, 3] == 1 and int((v != 0).sum()) == 1
+
+
+def test_repr():
+    dpf = DPF(device="cpu")
+    assert "uninitialized" in repr(dpf)
+    dpf.eval_init(torch.zeros((128, 4), dtype=torch.int32))
+    assert "entries=128" in repr(dpf) and "entry_size=4" in repr(dpf)
