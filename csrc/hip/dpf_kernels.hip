@@ -2,28 +2,39 @@
 //
 // Design (MI355X-first, not a port — see SURVEY.md §7):
 //   * One key per workgroup of Z = 1<<zlog threads (Z = 256 = 4 wave64 for
-//     n >= 512).  Grid = batch, so a 512-key batch puts 2 workgroups on
+//     n >= 512).  Grid = batch, so a 512-key batch puts >=2 workgroups on
 //     each of the 256 CUs.
 //   * Phase 1: breadth-first expansion of the GGM root to Z frontier seeds
 //     through an LDS ping-pong (zlog levels, ~2Z PRFs — negligible).
-//   * Phase 2: each thread owns ONE subtree and walks it with a sibling-stack
-//     DFS kept in LDS (stack index is wave-uniform per step, so ds_read/
-//     ds_write are conflict-free b128 ops).  All threads execute the same
-//     DFS schedule => zero divergence.  Each visited interior node expands
-//     BOTH children from one parent (for AES this shares one key schedule
-//     across the two encryptions — the reference re-expands per call,
-//     dpf_gpu/prf/prf.cu:159-184, flagged in its own TODO dpf.py:32-33).
+//   * Phase 2: each thread owns ONE subtree and walks it with a sibling-
+//     stack DFS.  All threads execute the same DFS schedule => zero
+//     divergence, and the stack index is wave-uniform per step so LDS
+//     accesses are conflict-free b128 ops.  The two HOT stack levels
+//     (popped on 3/4 of all steps) live in registers; the cold remainder
+//     shares the phase-1 ping-pong LDS region (they are never live at the
+//     same time), keeping LDS small enough for 3-4 workgroups per CU.
+//   * Each visited interior node expands BOTH children from one parent;
+//     for AES this shares one key schedule across the two encryptions
+//     (the reference re-expands per call: dpf_gpu/prf/prf.cu:159-184,
+//     flagged in its own TODO dpf.py:32-33).
 //   * Leaves: only the low 32 bits of a leaf share contribute to the
 //     (mod 2^32-truncated) output, because truncation is a ring hom of
 //     Z_2^128 -> Z_2^32.  The fused MAC is therefore 16 v_mad_u32 per leaf
 //     against u32 table rows — 16x less arithmetic and 4x less table
 //     traffic than the reference's mod-2^128 MAC (dpf_hybrid.cu:166-172),
-//     with bit-identical output.
+//     with bit-identical output.  Table loads are issued BEFORE the leaf
+//     PRFs of the same step, so ~1000 cycles of cipher work hides them.
+//   * AES T-table: a single 256-entry table replicated 32-way and
+//     interleaved (entry*32 + lane%32) so every lane reads its own LDS
+//     bank — measured 4.3x bank-conflict amplification with flat tables
+//     (SQ_LDS_BANK_CONFLICT 3.5e10 vs SQ_INSTS_LDS 8.1e9, profiles/).
+//     te1..te3 are byte rotations of te0 (1 v_alignbit per lookup) and
+//     sbox[x] = byte2 of te0[x], so one table serves the whole cipher.
 //   * Table layout: row(idx) = j<<(zlog+1) | t<<1 | b (leaf_perm in
-//     csrc/core/dpf_core.cc), so at DFS step j the workgroup reads one
-//     contiguous 32 KB slab and every lane reads its two rows as 8
-//     contiguous dwordx4 loads.  All workgroups stream the table in the
-//     same order => cross-key reuse in L2/LLC.
+//     csrc/core/dpf_core.cc): at DFS step j the workgroup reads one
+//     contiguous 32 KB slab; every lane reads its two rows as 8 contiguous
+//     dwordx4 loads.  All workgroups stream the table in the same order
+//     => cross-key reuse in L2/LLC.
 //
 // PRF device implementations mirror csrc/core/prf.cc bit-exactly (tested
 // by tests/test_gpu.py against the CPU core).
@@ -63,21 +74,26 @@ __device__ __forceinline__ uint4 add128(uint4 a, uint4 b) {
 __device__ __forceinline__ u32 rotl(u32 v, int s) {
   return (v << s) | (v >> (32 - s));  // lowers to v_alignbit_b32
 }
+__device__ __forceinline__ u32 rotr8(u32 v) { return (v >> 8) | (v << 24); }
+__device__ __forceinline__ u32 rotr16(u32 v) { return (v >> 16) | (v << 16); }
+__device__ __forceinline__ u32 rotr24(u32 v) { return (v >> 24) | (v << 8); }
 
-// ---------------------------------------------------------------------------
-// PRF constants / ids (wire values match the reference: prf.cu:6-9)
-// ---------------------------------------------------------------------------
 #define PRF_DUMMY 0
 #define PRF_SALSA20 1
 #define PRF_CHACHA20 2
 #define PRF_AES128 3
 
+// AES replicated-table geometry
+#define AES_REP 32
+#define AES_LDS_WORDS (256 * AES_REP)  // 32 KiB
+
 // ---------------------------------------------------------------------------
 // DUMMY: seed*(pos+4242) + (pos+4242) over Z_2^128
 // ---------------------------------------------------------------------------
 __device__ __forceinline__ uint4 prf_dummy_full(uint4 seed, u32 pos) {
-  unsigned __int128 s = ((unsigned __int128)(((u64)seed.w << 32) | seed.z) << 64) |
-                        (((u64)seed.y << 32) | seed.x);
+  unsigned __int128 s =
+      ((unsigned __int128)(((u64)seed.w << 32) | seed.z) << 64) |
+      (((u64)seed.y << 32) | seed.x);
   unsigned __int128 m = (unsigned __int128)(pos + 4242u);
   unsigned __int128 r = s * m + m;
   u64 lo = (u64)r, hi = (u64)(r >> 64);
@@ -94,7 +110,6 @@ __device__ __forceinline__ uint4 prf_dummy_full(uint4 seed, u32 pos) {
   d ^= rotl(c + b, 13);       \
   a ^= rotl(d + c, 18)
 
-// Runs the 12 rounds; returns the four result words (x[1..4]+in[1..4]).
 __device__ __forceinline__ uint4 salsa12_core(uint4 seed, u32 pos) {
   const u32 c0 = 0x65787061u, c5 = 0x6e642033u, c10 = 0x322d6279u,
             c15 = 0x7465206bu;
@@ -112,8 +127,41 @@ __device__ __forceinline__ uint4 salsa12_core(uint4 seed, u32 pos) {
     SALSA_QR(x10, x11, x8, x9);
     SALSA_QR(x15, x12, x13, x14);
   }
-  // result words high->low: out1, out2, out3, out4
   return make_uint4(x4 + seed.x, x3 + seed.y, x2 + seed.z, x1 + seed.w);
+}
+
+// Both children interleaved in one pass: the two blocks (pos=0 / pos=1)
+// are independent dependency chains, so interleaving doubles the ILP the
+// SIMD can draw on — the DFS runs at ~2 waves/SIMD where a single chain's
+// QR latency is not fully hidden.
+#define SALSA_QR2(a, b, c, d, a2, b2, c2, d2) \
+  b ^= rotl(a + d, 7);   b2 ^= rotl(a2 + d2, 7);   \
+  c ^= rotl(b + a, 9);   c2 ^= rotl(b2 + a2, 9);   \
+  d ^= rotl(c + b, 13);  d2 ^= rotl(c2 + b2, 13);  \
+  a ^= rotl(d + c, 18);  a2 ^= rotl(d2 + c2, 18)
+
+__device__ __forceinline__ void salsa12_pair_core(uint4 seed, uint4& r0,
+                                                  uint4& r1) {
+  const u32 c0 = 0x65787061u, c5 = 0x6e642033u, c10 = 0x322d6279u,
+            c15 = 0x7465206bu;
+  u32 x0 = c0, x1 = seed.w, x2 = seed.z, x3 = seed.y, x4 = seed.x, x5 = c5,
+      x6 = 0, x7 = 0, x8 = 0, x9 = 0, x10 = c10, x11 = 0, x12 = 0, x13 = 0,
+      x14 = 0, x15 = c15;
+  u32 y0 = c0, y1 = x1, y2 = x2, y3 = x3, y4 = x4, y5 = c5, y6 = 0, y7 = 0,
+      y8 = 0, y9 = 1, y10 = c10, y11 = 0, y12 = 0, y13 = 0, y14 = 0, y15 = c15;
+#pragma unroll
+  for (int r = 0; r < 6; ++r) {
+    SALSA_QR2(x0, x4, x8, x12, y0, y4, y8, y12);
+    SALSA_QR2(x5, x9, x13, x1, y5, y9, y13, y1);
+    SALSA_QR2(x10, x14, x2, x6, y10, y14, y2, y6);
+    SALSA_QR2(x15, x3, x7, x11, y15, y3, y7, y11);
+    SALSA_QR2(x0, x1, x2, x3, y0, y1, y2, y3);
+    SALSA_QR2(x5, x6, x7, x4, y5, y6, y7, y4);
+    SALSA_QR2(x10, x11, x8, x9, y10, y11, y8, y9);
+    SALSA_QR2(x15, x12, x13, x14, y15, y12, y13, y14);
+  }
+  r0 = make_uint4(x4 + seed.x, x3 + seed.y, x2 + seed.z, x1 + seed.w);
+  r1 = make_uint4(y4 + seed.x, y3 + seed.y, y2 + seed.z, y1 + seed.w);
 }
 
 // ---------------------------------------------------------------------------
@@ -143,13 +191,449 @@ __device__ __forceinline__ uint4 chacha12_core(uint4 seed, u32 pos) {
   return make_uint4(x7 + seed.x, x6 + seed.y, x5 + seed.z, x4 + seed.w);
 }
 
+#define CHACHA_QR2(a, b, c, d, a2, b2, c2, d2)                      \
+  a += b;  a2 += b2;  d ^= a;  d2 ^= a2;                            \
+  d = rotl(d, 16);  d2 = rotl(d2, 16);                              \
+  c += d;  c2 += d2;  b ^= c;  b2 ^= c2;                            \
+  b = rotl(b, 12);  b2 = rotl(b2, 12);                              \
+  a += b;  a2 += b2;  d ^= a;  d2 ^= a2;                            \
+  d = rotl(d, 8);  d2 = rotl(d2, 8);                                \
+  c += d;  c2 += d2;  b ^= c;  b2 ^= c2;                            \
+  b = rotl(b, 7);  b2 = rotl(b2, 7)
+
+__device__ __forceinline__ void chacha12_pair_core(uint4 seed, uint4& r0,
+                                                   uint4& r1) {
+  const u32 k0 = 0x65787061u, k1 = 0x6e642033u, k2 = 0x322d6279u,
+            k3 = 0x7465206bu;
+  u32 x0 = k0, x1 = k1, x2 = k2, x3 = k3;
+  u32 x4 = seed.w, x5 = seed.z, x6 = seed.y, x7 = seed.x;
+  u32 x8 = 0, x9 = 0, x10 = 0, x11 = 0, x12 = 0, x13 = 0, x14 = 0, x15 = 0;
+  u32 y0 = k0, y1 = k1, y2 = k2, y3 = k3;
+  u32 y4 = x4, y5 = x5, y6 = x6, y7 = x7;
+  u32 y8 = 0, y9 = 0, y10 = 0, y11 = 0, y12 = 0, y13 = 1, y14 = 0, y15 = 0;
+#pragma unroll
+  for (int r = 0; r < 6; ++r) {
+    CHACHA_QR2(x0, x4, x8, x12, y0, y4, y8, y12);
+    CHACHA_QR2(x1, x5, x9, x13, y1, y5, y9, y13);
+    CHACHA_QR2(x2, x6, x10, x14, y2, y6, y10, y14);
+    CHACHA_QR2(x3, x7, x11, x15, y3, y7, y11, y15);
+    CHACHA_QR2(x0, x5, x10, x15, y0, y5, y10, y15);
+    CHACHA_QR2(x1, x6, x11, x12, y1, y6, y11, y12);
+    CHACHA_QR2(x2, x7, x8, x13, y2, y7, y8, y13);
+    CHACHA_QR2(x3, x4, x9, x14, y3, y4, y9, y14);
+  }
+  r0 = make_uint4(x7 + seed.x, x6 + seed.y, x5 + seed.z, x4 + seed.w);
+  r1 = make_uint4(y7 + seed.x, y6 + seed.y, y5 + seed.z, y4 + seed.w);
+}
+
 // ---------------------------------------------------------------------------
-// AES-128 with LDS T-tables.  aes_lds layout: te0|te1|te2|te3|sbox, each
-// 256 u32 (tables generated host-side by gpudpf::aes128_tables).
-// Key schedule is expanded once per NODE and shared by both children.
+// AES-128, replicated-LDS variant (fused kernel).  aes_lds holds te0
+// replicated AES_REP-way, interleaved: word(entry e, copy c) at e*32+c.
+// Lane l always uses copy l%32 => bank (addr/4)%32 == l%32: conflict-free.
+//   te0[x] bytes (MSB..LSB) = [2s, s, s, 3s];  te1 = ror8(te0),
+//   te2 = ror16, te3 = ror24;  sbox[x] = byte2 of te0[x].
 // ---------------------------------------------------------------------------
-__device__ __forceinline__ void aes_expand_rk(uint4 seed, const u32* sb,
+struct AesLds {
+  const u32* rep;  // replicated te0
+  u32 lane;        // threadIdx.x % 32
+  __device__ __forceinline__ u32 te0(u32 x) const { return rep[x * AES_REP + lane]; }
+  __device__ __forceinline__ u32 sbox(u32 x) const { return (te0(x) >> 16) & 0xff; }
+};
+
+__device__ __forceinline__ void aes_expand_rk(uint4 seed, const AesLds& T,
                                               u32 rk[44]) {
+  rk[0] = __builtin_bswap32(seed.x);
+  rk[1] = __builtin_bswap32(seed.y);
+  rk[2] = __builtin_bswap32(seed.z);
+  rk[3] = __builtin_bswap32(seed.w);
+  u32 rcon = 0x01u;
+#pragma unroll
+  for (int r = 1; r <= 10; ++r) {
+    u32 w = rk[4 * r - 1];
+    w = (w << 8) | (w >> 24);
+    w = (T.sbox((w >> 24) & 0xff) << 24) | (T.sbox((w >> 16) & 0xff) << 16) |
+        (T.sbox((w >> 8) & 0xff) << 8) | T.sbox(w & 0xff);
+    w ^= (rcon << 24);
+    rcon = (rcon << 1) ^ ((rcon & 0x80u) ? 0x11bu : 0u);
+    rcon &= 0xffu;
+    rk[4 * r] = rk[4 * r - 4] ^ w;
+    rk[4 * r + 1] = rk[4 * r - 3] ^ rk[4 * r];
+    rk[4 * r + 2] = rk[4 * r - 2] ^ rk[4 * r + 1];
+    rk[4 * r + 3] = rk[4 * r - 1] ^ rk[4 * r + 2];
+  }
+}
+
+// LOW=true computes only the low 32 bits of the result (4 final-round
+// lookups instead of 16).
+template <bool LOW>
+__device__ __forceinline__ uint4 aes_cipher_rep(const u32 rk[44],
+                                                const AesLds& T, u32 pos) {
+  u32 s0 = (pos << 24) ^ rk[0], s1 = rk[1], s2 = rk[2], s3 = rk[3];
+#pragma unroll
+  for (int r = 1; r < 10; ++r) {
+    u32 n0 = T.te0(s0 >> 24) ^ rotr8(T.te0((s1 >> 16) & 0xff)) ^
+             rotr16(T.te0((s2 >> 8) & 0xff)) ^ rotr24(T.te0(s3 & 0xff)) ^
+             rk[4 * r];
+    u32 n1 = T.te0(s1 >> 24) ^ rotr8(T.te0((s2 >> 16) & 0xff)) ^
+             rotr16(T.te0((s3 >> 8) & 0xff)) ^ rotr24(T.te0(s0 & 0xff)) ^
+             rk[4 * r + 1];
+    u32 n2 = T.te0(s2 >> 24) ^ rotr8(T.te0((s3 >> 16) & 0xff)) ^
+             rotr16(T.te0((s0 >> 8) & 0xff)) ^ rotr24(T.te0(s1 & 0xff)) ^
+             rk[4 * r + 2];
+    u32 n3 = T.te0(s3 >> 24) ^ rotr8(T.te0((s0 >> 16) & 0xff)) ^
+             rotr16(T.te0((s1 >> 8) & 0xff)) ^ rotr24(T.te0(s2 & 0xff)) ^
+             rk[4 * r + 3];
+    s0 = n0; s1 = n1; s2 = n2; s3 = n3;
+  }
+  u32 o0 = ((T.sbox(s0 >> 24) << 24) | (T.sbox((s1 >> 16) & 0xff) << 16) |
+            (T.sbox((s2 >> 8) & 0xff) << 8) | T.sbox(s3 & 0xff)) ^ rk[40];
+  if constexpr (LOW) return make_uint4(__builtin_bswap32(o0), 0, 0, 0);
+  u32 o1 = ((T.sbox(s1 >> 24) << 24) | (T.sbox((s2 >> 16) & 0xff) << 16) |
+            (T.sbox((s3 >> 8) & 0xff) << 8) | T.sbox(s0 & 0xff)) ^ rk[41];
+  u32 o2 = ((T.sbox(s2 >> 24) << 24) | (T.sbox((s3 >> 16) & 0xff) << 16) |
+            (T.sbox((s0 >> 8) & 0xff) << 8) | T.sbox(s1 & 0xff)) ^ rk[42];
+  u32 o3 = ((T.sbox(s3 >> 24) << 24) | (T.sbox((s0 >> 16) & 0xff) << 16) |
+            (T.sbox((s1 >> 8) & 0xff) << 8) | T.sbox(s2 & 0xff)) ^ rk[43];
+  return make_uint4(__builtin_bswap32(o0), __builtin_bswap32(o1),
+                    __builtin_bswap32(o2), __builtin_bswap32(o3));
+}
+
+// Both children (pos 0 and pos 1) under one key schedule, rounds
+// interleaved for 2x ILP on both the LDS and VALU pipes.
+template <bool LOW>
+__device__ __forceinline__ void aes_cipher_pair(const u32 rk[44],
+                                                const AesLds& T, uint4& ra,
+                                                uint4& rb) {
+  u32 s0 = rk[0], s1 = rk[1], s2 = rk[2], s3 = rk[3];
+  u32 u0 = (1u << 24) ^ rk[0], u1 = rk[1], u2 = rk[2], u3 = rk[3];
+#pragma unroll
+  for (int r = 1; r < 10; ++r) {
+    u32 n0 = T.te0(s0 >> 24) ^ rotr8(T.te0((s1 >> 16) & 0xff)) ^
+             rotr16(T.te0((s2 >> 8) & 0xff)) ^ rotr24(T.te0(s3 & 0xff)) ^
+             rk[4 * r];
+    u32 m0 = T.te0(u0 >> 24) ^ rotr8(T.te0((u1 >> 16) & 0xff)) ^
+             rotr16(T.te0((u2 >> 8) & 0xff)) ^ rotr24(T.te0(u3 & 0xff)) ^
+             rk[4 * r];
+    u32 n1 = T.te0(s1 >> 24) ^ rotr8(T.te0((s2 >> 16) & 0xff)) ^
+             rotr16(T.te0((s3 >> 8) & 0xff)) ^ rotr24(T.te0(s0 & 0xff)) ^
+             rk[4 * r + 1];
+    u32 m1 = T.te0(u1 >> 24) ^ rotr8(T.te0((u2 >> 16) & 0xff)) ^
+             rotr16(T.te0((u3 >> 8) & 0xff)) ^ rotr24(T.te0(u0 & 0xff)) ^
+             rk[4 * r + 1];
+    u32 n2 = T.te0(s2 >> 24) ^ rotr8(T.te0((s3 >> 16) & 0xff)) ^
+             rotr16(T.te0((s0 >> 8) & 0xff)) ^ rotr24(T.te0(s1 & 0xff)) ^
+             rk[4 * r + 2];
+    u32 m2 = T.te0(u2 >> 24) ^ rotr8(T.te0((u3 >> 16) & 0xff)) ^
+             rotr16(T.te0((u0 >> 8) & 0xff)) ^ rotr24(T.te0(u1 & 0xff)) ^
+             rk[4 * r + 2];
+    u32 n3 = T.te0(s3 >> 24) ^ rotr8(T.te0((s0 >> 16) & 0xff)) ^
+             rotr16(T.te0((s1 >> 8) & 0xff)) ^ rotr24(T.te0(s2 & 0xff)) ^
+             rk[4 * r + 3];
+    u32 m3 = T.te0(u3 >> 24) ^ rotr8(T.te0((u0 >> 16) & 0xff)) ^
+             rotr16(T.te0((u1 >> 8) & 0xff)) ^ rotr24(T.te0(u2 & 0xff)) ^
+             rk[4 * r + 3];
+    s0 = n0; s1 = n1; s2 = n2; s3 = n3;
+    u0 = m0; u1 = m1; u2 = m2; u3 = m3;
+  }
+  u32 o0 = ((T.sbox(s0 >> 24) << 24) | (T.sbox((s1 >> 16) & 0xff) << 16) |
+            (T.sbox((s2 >> 8) & 0xff) << 8) | T.sbox(s3 & 0xff)) ^ rk[40];
+  u32 p0 = ((T.sbox(u0 >> 24) << 24) | (T.sbox((u1 >> 16) & 0xff) << 16) |
+            (T.sbox((u2 >> 8) & 0xff) << 8) | T.sbox(u3 & 0xff)) ^ rk[40];
+  if constexpr (LOW) {
+    ra = make_uint4(__builtin_bswap32(o0), 0, 0, 0);
+    rb = make_uint4(__builtin_bswap32(p0), 0, 0, 0);
+    return;
+  }
+  u32 o1 = ((T.sbox(s1 >> 24) << 24) | (T.sbox((s2 >> 16) & 0xff) << 16) |
+            (T.sbox((s3 >> 8) & 0xff) << 8) | T.sbox(s0 & 0xff)) ^ rk[41];
+  u32 p1 = ((T.sbox(u1 >> 24) << 24) | (T.sbox((u2 >> 16) & 0xff) << 16) |
+            (T.sbox((u3 >> 8) & 0xff) << 8) | T.sbox(u0 & 0xff)) ^ rk[41];
+  u32 o2 = ((T.sbox(s2 >> 24) << 24) | (T.sbox((s3 >> 16) & 0xff) << 16) |
+            (T.sbox((s0 >> 8) & 0xff) << 8) | T.sbox(s1 & 0xff)) ^ rk[42];
+  u32 p2 = ((T.sbox(u2 >> 24) << 24) | (T.sbox((u3 >> 16) & 0xff) << 16) |
+            (T.sbox((u0 >> 8) & 0xff) << 8) | T.sbox(u1 & 0xff)) ^ rk[42];
+  u32 o3 = ((T.sbox(s3 >> 24) << 24) | (T.sbox((s0 >> 16) & 0xff) << 16) |
+            (T.sbox((s1 >> 8) & 0xff) << 8) | T.sbox(s2 & 0xff)) ^ rk[43];
+  u32 p3 = ((T.sbox(u3 >> 24) << 24) | (T.sbox((u0 >> 16) & 0xff) << 16) |
+            (T.sbox((u1 >> 8) & 0xff) << 8) | T.sbox(u2 & 0xff)) ^ rk[43];
+  ra = make_uint4(__builtin_bswap32(o0), __builtin_bswap32(o1),
+                  __builtin_bswap32(o2), __builtin_bswap32(o3));
+  rb = make_uint4(__builtin_bswap32(p0), __builtin_bswap32(p1),
+                  __builtin_bswap32(p2), __builtin_bswap32(p3));
+}
+
+// ---------------------------------------------------------------------------
+// PRF dispatch (fused kernel: AES uses the replicated LDS table)
+// ---------------------------------------------------------------------------
+template <int PRF>
+__device__ __forceinline__ uint4 prf_full(uint4 seed, u32 pos,
+                                          const AesLds& T) {
+  if constexpr (PRF == PRF_DUMMY) return prf_dummy_full(seed, pos);
+  if constexpr (PRF == PRF_SALSA20) return salsa12_core(seed, pos);
+  if constexpr (PRF == PRF_CHACHA20) return chacha12_core(seed, pos);
+  if constexpr (PRF == PRF_AES128) {
+    u32 rk[44];
+    aes_expand_rk(seed, T, rk);
+    return aes_cipher_rep<false>(rk, T, pos);
+  }
+}
+
+template <int PRF>
+__device__ __forceinline__ void prf_pair(uint4 seed, const AesLds& T,
+                                         uint4& r0, uint4& r1) {
+  if constexpr (PRF == PRF_AES128) {
+    u32 rk[44];
+    aes_expand_rk(seed, T, rk);
+    aes_cipher_pair<false>(rk, T, r0, r1);
+  } else if constexpr (PRF == PRF_SALSA20) {
+    salsa12_pair_core(seed, r0, r1);
+  } else if constexpr (PRF == PRF_CHACHA20) {
+    chacha12_pair_core(seed, r0, r1);
+  } else {
+    r0 = prf_full<PRF>(seed, 0, T);
+    r1 = prf_full<PRF>(seed, 1, T);
+  }
+}
+
+template <int PRF>
+__device__ __forceinline__ void prf_pair_low(uint4 seed, const AesLds& T,
+                                             u32& r0, u32& r1) {
+  if constexpr (PRF == PRF_DUMMY) {
+    r0 = seed.x * 4242u + 4242u;
+    r1 = seed.x * 4243u + 4243u;
+  } else if constexpr (PRF == PRF_SALSA20) {
+    uint4 a, b;
+    salsa12_pair_core(seed, a, b);
+    r0 = a.x;
+    r1 = b.x;
+  } else if constexpr (PRF == PRF_CHACHA20) {
+    uint4 a, b;
+    chacha12_pair_core(seed, a, b);
+    r0 = a.x;
+    r1 = b.x;
+  } else {
+    u32 rk[44];
+    aes_expand_rk(seed, T, rk);
+    uint4 a, b;
+    aes_cipher_pair<true>(rk, T, a, b);
+    r0 = a.x;
+    r1 = b.x;
+  }
+}
+
+template <int PRF>
+__device__ __forceinline__ void expand_pair(uint4 seed, int i,
+                                            const uint4* cw_lds,
+                                            const AesLds& T, uint4& c0,
+                                            uint4& c1) {
+  uint4 p0, p1;
+  prf_pair<PRF>(seed, T, p0, p1);
+  const int sel = (int)(seed.x & 1u);
+  c0 = add128(p0, cw_lds[sel * 64 + i * 2 + 0]);
+  c1 = add128(p1, cw_lds[sel * 64 + i * 2 + 1]);
+}
+
+template <int PRF>
+__device__ __forceinline__ void expand_leaf_low(uint4 seed,
+                                                const uint4* cw_lds,
+                                                const AesLds& T, u32& v0,
+                                                u32& v1) {
+  u32 p0, p1;
+  prf_pair_low<PRF>(seed, T, p0, p1);
+  const int sel = (int)(seed.x & 1u);
+  v0 = p0 + cw_lds[sel * 64 + 0].x;
+  v1 = p1 + cw_lds[sel * 64 + 1].x;
+}
+
+// ---------------------------------------------------------------------------
+// Main kernel.  LDS map (u32 granularity):
+//   [cw: 128 uint4][shared: Z*max(2, DS-3) uint4][aes: 0/8192 u32][red]
+// The `shared` region is the phase-1 ping-pong (2*Z uint4) first, then the
+// cold DFS stack (levels 1..DS-3) — never live simultaneously.
+// Hot stack levels DS-1 / DS-2 are the registers rtop1 / rtop2.
+// ---------------------------------------------------------------------------
+// slog: log2 of the DFS range split — each key's 2^(DS-1) leaf pairs are
+// divided over 2^slog workgroups (j-split).  The split changes neither the
+// table layout nor the per-thread subtree assignment: workgroup (key, seg)
+// walks pairs [seg*P, (seg+1)*P) of every thread's subtree, paying one
+// extra targeted descent (DS-1 pair expansions) to enter at seg*P.  Fused
+// partials combine with wrapping u32 atomicAdd (exact mod 2^32).  This is
+// what fills 256 CUs at small batch (batch 512 alone = only 2 WGs/CU) and
+// doubles as the single-key low-latency mode (the reference needs a
+// separate cooperative-groups kernel for that: dpf_coop.cu).
+template <int PRF, bool FUSED>
+__global__ __launch_bounds__(256) void dpf_eval_kernel(
+    const int* __restrict__ keys, const u32* __restrict__ table,
+    u32* __restrict__ out, const u32* __restrict__ aes_tabs, int depth,
+    int zlog, int slog, long long n) {
+  extern __shared__ u32 smem[];
+  const int Z = 1 << zlog;
+  const int DS = depth - zlog;  // subtree splits per thread (>= 1)
+  const int lds_levels = DS > 3 ? DS - 3 : 0;
+  const int t = (int)threadIdx.x;
+  const int key_id = (int)(blockIdx.x >> slog);
+  const int seg = (int)(blockIdx.x & ((1u << slog) - 1));
+  const long long key_base = (long long)key_id * 524;
+
+  uint4* cw_lds = reinterpret_cast<uint4*>(smem);   // 128 entries
+  uint4* shared_region = cw_lds + 128;              // Z*max(2, DS-3)
+  const int shared_u4 = Z * (lds_levels > 2 ? lds_levels : 2);
+  u32* aes_lds = reinterpret_cast<u32*>(shared_region + shared_u4);
+  u32* red = aes_lds + (PRF == PRF_AES128 ? AES_LDS_WORDS : 0);
+
+  // Stage codewords; replicate the AES te0 table 32-way interleaved.
+  for (int idx = t; idx < 128; idx += blockDim.x)
+    cw_lds[idx] = reinterpret_cast<const uint4*>(keys + key_base + 4)[idx];
+  if constexpr (PRF == PRF_AES128) {
+    for (int e = t; e < 256; e += blockDim.x) {
+      const u32 v = aes_tabs[e];  // global layout: te0 at offset 0
+#pragma unroll
+      for (int c = 0; c < AES_REP; ++c) aes_lds[e * AES_REP + c] = v;
+    }
+  }
+  AesLds T{aes_lds, (u32)(t & 31)};
+  uint4* pp = shared_region;
+  if (t == 0) {
+    const int* rp = keys + key_base + 516;
+    pp[0] = make_uint4((u32)rp[0], (u32)rp[1], (u32)rp[2], (u32)rp[3]);
+  }
+  __syncthreads();
+
+  // Phase 1: root -> Z frontier seeds (frontier position t has the
+  // first-consumed index bit as its MSB: t = bitrev(idx & (Z-1))).
+  uint4* a = pp;
+  uint4* b = pp + Z;
+  for (int l = 1; l <= zlog; ++l) {
+    const int i = depth - l;
+    if (t < (1 << l)) {
+      uint4 parent = a[t >> 1];
+      uint4 v = prf_full<PRF>(parent, (u32)(t & 1), T);
+      const int sel = (int)(parent.x & 1u);
+      b[t] = add128(v, cw_lds[sel * 64 + i * 2 + (t & 1)]);
+    }
+    __syncthreads();
+    uint4* tmp = a;
+    a = b;
+    b = tmp;
+  }
+  uint4 cur = a[t];
+  __syncthreads();  // everyone holds their frontier seed; pp is now free
+  uint4* stack = shared_region;  // cold stack levels 1..DS-3 (slot d-1)
+
+  // Targeted descent to leaf-pair j_lo: at split d take bit (DS-1-d) of
+  // j_lo, always recording the bit-1 child in the level's sibling slot
+  // (when the bit is 1 the slot is stale, but it is provably rewritten by
+  // a later descent before its next pop).
+  const long long pairs = 1LL << (DS - 1);
+  const long long j_lo = (long long)seg * (pairs >> slog);
+  const long long j_hi = j_lo + (pairs >> slog);
+  uint4 rtop1 = cur, rtop2 = cur;  // hot levels DS-1 / DS-2
+  for (int d = 1; d <= DS - 1; ++d) {
+    uint4 c0, c1;
+    expand_pair<PRF>(cur, DS - d, cw_lds, T, c0, c1);
+    if (d == DS - 1) rtop1 = c1;
+    else if (d == DS - 2) rtop2 = c1;
+    else stack[(d - 1) * Z + t] = c1;
+    cur = ((j_lo >> (DS - 1 - d)) & 1) ? c1 : c0;
+  }
+
+  u32 acc[16];
+#pragma unroll
+  for (int w = 0; w < 16; ++w) acc[w] = 0;
+
+  for (long long j = j_lo; j < j_hi; ++j) {
+    // Issue the two table-row loads first: the leaf ciphers below hide
+    // their latency.
+    uint4 r0q[4], r1q[4];
+    if constexpr (FUSED) {
+      const uint4* rows =
+          reinterpret_cast<const uint4*>(table + ((j << (zlog + 1)) +
+                                                  ((long long)t << 1)) * 16);
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        r0q[q] = rows[q];
+        r1q[q] = rows[q + 4];
+      }
+    }
+
+    u32 v0, v1;
+    expand_leaf_low<PRF>(cur, cw_lds, T, v0, v1);
+
+    if constexpr (FUSED) {
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        acc[4 * q + 0] += v0 * r0q[q].x + v1 * r1q[q].x;
+        acc[4 * q + 1] += v0 * r0q[q].y + v1 * r1q[q].y;
+        acc[4 * q + 2] += v0 * r0q[q].z + v1 * r1q[q].z;
+        acc[4 * q + 3] += v0 * r0q[q].w + v1 * r1q[q].w;
+      }
+    } else {
+      u32* orow = out + (u64)key_id * (u64)n +
+                  (u64)((j << (zlog + 1)) + ((long long)t << 1));
+      orow[0] = v0;
+      orow[1] = v1;
+    }
+
+    if (j + 1 == j_hi) break;
+    // Pop the deepest pending sibling (register-resident on 3/4 of steps)
+    // and descend its bit-0 spine.
+    const int c = (int)(__ffsll((unsigned long long)(j + 1)) - 1);
+    if (c == 0) {
+      cur = rtop1;
+      continue;  // next leaf-parent is the sibling itself
+    }
+    const int dpop = DS - 1 - c;
+    cur = (c == 1) ? rtop2 : stack[(dpop - 1) * Z + t];
+    for (int d = dpop + 1; d <= DS - 1; ++d) {
+      uint4 c0, c1;
+      expand_pair<PRF>(cur, DS - d, cw_lds, T, c0, c1);
+      if (d == DS - 1) rtop1 = c1;
+      else if (d == DS - 2) rtop2 = c1;
+      else stack[(d - 1) * Z + t] = c1;
+      cur = c0;
+    }
+  }
+
+  if constexpr (FUSED) {
+    // Wave shfl reduction, then cross-wave sum through LDS.
+#pragma unroll
+    for (int w = 0; w < 16; ++w) {
+      u32 v = acc[w];
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+      acc[w] = v;
+    }
+    const int lane = t & 63, wave = t >> 6, nwaves = (int)blockDim.x >> 6;
+    __syncthreads();  // stack region re-used as reduction scratch is NOT —
+                      // red is separate; this sync orders DFS completion
+    if (lane == 0) {
+#pragma unroll
+      for (int w = 0; w < 16; ++w) red[wave * 16 + w] = acc[w];
+    }
+    __syncthreads();
+    if (t < 16) {
+      u32 s = 0;
+      for (int wv = 0; wv < nwaves; ++wv) s += red[wv * 16 + t];
+      // Wrapping u32 atomic add combines the 2^slog segment partials
+      // exactly (mod 2^32); `out` is zeroed by the caller.
+      atomicAdd(&out[(u64)key_id * 16 + t], s);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Naive oracle kernel: EvaluateFlat per (key, leaf), natural order.
+// (Test-only; uses the flat 4-table AES layout at aes_tabs[0..1279]:
+// te0|te1|te2|te3|sbox.)
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ uint4 aes_flat(const u32* tabs, uint4 seed,
+                                          u32 pos) {
+  const u32* te0 = tabs;
+  const u32* te1 = tabs + 256;
+  const u32* te2 = tabs + 512;
+  const u32* te3 = tabs + 768;
+  const u32* sb = tabs + 1024;
+  u32 rk[44];
   rk[0] = __builtin_bswap32(seed.x);
   rk[1] = __builtin_bswap32(seed.y);
   rk[2] = __builtin_bswap32(seed.z);
@@ -169,16 +653,6 @@ __device__ __forceinline__ void aes_expand_rk(uint4 seed, const u32* sb,
     rk[4 * r + 2] = rk[4 * r - 2] ^ rk[4 * r + 1];
     rk[4 * r + 3] = rk[4 * r - 1] ^ rk[4 * r + 2];
   }
-}
-
-// Encrypt the 1-byte counter block (pos in {0,1}); full 128-bit result.
-__device__ __forceinline__ uint4 aes_cipher(const u32 rk[44], const u32* tabs,
-                                            u32 pos) {
-  const u32* te0 = tabs;
-  const u32* te1 = tabs + 256;
-  const u32* te2 = tabs + 512;
-  const u32* te3 = tabs + 768;
-  const u32* sb = tabs + 1024;
   u32 s0 = (pos << 24) ^ rk[0], s1 = rk[1], s2 = rk[2], s3 = rk[3];
 #pragma unroll
   for (int r = 1; r < 10; ++r) {
@@ -200,246 +674,23 @@ __device__ __forceinline__ uint4 aes_cipher(const u32 rk[44], const u32* tabs,
             (sb[(s0 >> 8) & 0xff] << 8) | sb[s1 & 0xff]) ^ rk[42];
   u32 o3 = ((sb[s3 >> 24] << 24) | (sb[(s0 >> 16) & 0xff] << 16) |
             (sb[(s1 >> 8) & 0xff] << 8) | sb[s2 & 0xff]) ^ rk[43];
-  // ciphertext bytes little-endian -> u128 words
   return make_uint4(__builtin_bswap32(o0), __builtin_bswap32(o1),
                     __builtin_bswap32(o2), __builtin_bswap32(o3));
 }
 
-// Low-32 result only (leaf fast path: result.x = bswap(o0) needs 4 final-
-// round sbox lookups instead of 16).
-__device__ __forceinline__ u32 aes_cipher_low(const u32 rk[44], const u32* tabs,
-                                              u32 pos) {
-  const u32* te0 = tabs;
-  const u32* te1 = tabs + 256;
-  const u32* te2 = tabs + 512;
-  const u32* te3 = tabs + 768;
-  const u32* sb = tabs + 1024;
-  u32 s0 = (pos << 24) ^ rk[0], s1 = rk[1], s2 = rk[2], s3 = rk[3];
-#pragma unroll
-  for (int r = 1; r < 10; ++r) {
-    u32 n0 = te0[s0 >> 24] ^ te1[(s1 >> 16) & 0xff] ^ te2[(s2 >> 8) & 0xff] ^
-             te3[s3 & 0xff] ^ rk[4 * r];
-    u32 n1 = te0[s1 >> 24] ^ te1[(s2 >> 16) & 0xff] ^ te2[(s3 >> 8) & 0xff] ^
-             te3[s0 & 0xff] ^ rk[4 * r + 1];
-    u32 n2 = te0[s2 >> 24] ^ te1[(s3 >> 16) & 0xff] ^ te2[(s0 >> 8) & 0xff] ^
-             te3[s1 & 0xff] ^ rk[4 * r + 2];
-    u32 n3 = te0[s3 >> 24] ^ te1[(s0 >> 16) & 0xff] ^ te2[(s1 >> 8) & 0xff] ^
-             te3[s2 & 0xff] ^ rk[4 * r + 3];
-    s0 = n0; s1 = n1; s2 = n2; s3 = n3;
-  }
-  u32 o0 = ((sb[s0 >> 24] << 24) | (sb[(s1 >> 16) & 0xff] << 16) |
-            (sb[(s2 >> 8) & 0xff] << 8) | sb[s3 & 0xff]) ^ rk[40];
-  return __builtin_bswap32(o0);
-}
-
-// ---------------------------------------------------------------------------
-// PRF dispatch: full single, full pair (shared AES key schedule), low pair
-// ---------------------------------------------------------------------------
 template <int PRF>
-__device__ __forceinline__ uint4 prf_full(uint4 seed, u32 pos, const u32* aes) {
+__device__ __forceinline__ uint4 prf_naive(const u32* aes_lds, uint4 seed,
+                                           u32 pos) {
   if constexpr (PRF == PRF_DUMMY) return prf_dummy_full(seed, pos);
   if constexpr (PRF == PRF_SALSA20) return salsa12_core(seed, pos);
   if constexpr (PRF == PRF_CHACHA20) return chacha12_core(seed, pos);
-  if constexpr (PRF == PRF_AES128) {
-    u32 rk[44];
-    aes_expand_rk(seed, aes + 1024, rk);
-    return aes_cipher(rk, aes, pos);
-  }
+  if constexpr (PRF == PRF_AES128) return aes_flat(aes_lds, seed, pos);
 }
 
 template <int PRF>
-__device__ __forceinline__ void prf_pair(uint4 seed, const u32* aes, uint4& r0,
-                                         uint4& r1) {
-  if constexpr (PRF == PRF_AES128) {
-    u32 rk[44];
-    aes_expand_rk(seed, aes + 1024, rk);
-    r0 = aes_cipher(rk, aes, 0);
-    r1 = aes_cipher(rk, aes, 1);
-  } else {
-    r0 = prf_full<PRF>(seed, 0, aes);
-    r1 = prf_full<PRF>(seed, 1, aes);
-  }
-}
-
-template <int PRF>
-__device__ __forceinline__ void prf_pair_low(uint4 seed, const u32* aes,
-                                             u32& r0, u32& r1) {
-  if constexpr (PRF == PRF_DUMMY) {
-    r0 = seed.x * 4242u + 4242u;
-    r1 = seed.x * 4243u + 4243u;
-  } else if constexpr (PRF == PRF_SALSA20) {
-    r0 = salsa12_core(seed, 0).x;
-    r1 = salsa12_core(seed, 1).x;
-  } else if constexpr (PRF == PRF_CHACHA20) {
-    r0 = chacha12_core(seed, 0).x;
-    r1 = chacha12_core(seed, 1).x;
-  } else {
-    u32 rk[44];
-    aes_expand_rk(seed, aes + 1024, rk);
-    r0 = aes_cipher_low(rk, aes, 0);
-    r1 = aes_cipher_low(rk, aes, 1);
-  }
-}
-
-// Expand both children of `seed` at eval-level i (adds correction words).
-template <int PRF>
-__device__ __forceinline__ void expand_pair(uint4 seed, int i,
-                                            const uint4* cw_lds, const u32* aes,
-                                            uint4& c0, uint4& c1) {
-  uint4 p0, p1;
-  prf_pair<PRF>(seed, aes, p0, p1);
-  const int sel = (int)(seed.x & 1u);
-  c0 = add128(p0, cw_lds[sel * 64 + i * 2 + 0]);
-  c1 = add128(p1, cw_lds[sel * 64 + i * 2 + 1]);
-}
-
-// Leaf expansion: low-32 values of both children at eval-level 0.
-template <int PRF>
-__device__ __forceinline__ void expand_leaf_low(uint4 seed, const uint4* cw_lds,
-                                                const u32* aes, u32& v0,
-                                                u32& v1) {
-  u32 p0, p1;
-  prf_pair_low<PRF>(seed, aes, p0, p1);
-  const int sel = (int)(seed.x & 1u);
-  v0 = p0 + cw_lds[sel * 64 + 0].x;
-  v1 = p1 + cw_lds[sel * 64 + 1].x;
-}
-
-// ---------------------------------------------------------------------------
-// Main kernel: phase-1 frontier + per-thread sibling-stack DFS.
-// FUSED=true: accumulate the table inner product; else write raw shares.
-// ---------------------------------------------------------------------------
-template <int PRF, bool FUSED>
-__global__ __launch_bounds__(256) void dpf_eval_kernel(const int* __restrict__ keys,
-                                const u32* __restrict__ table,
-                                u32* __restrict__ out,
-                                const u32* __restrict__ aes_tabs, int depth,
-                                int zlog, long long n) {
-  extern __shared__ u32 smem[];
-  const int Z = 1 << zlog;
-  const int DS = depth - zlog;  // subtree splits per thread (>= 1)
-  const int t = (int)threadIdx.x;
-  const long long key_base = (long long)blockIdx.x * 524;
-
-  uint4* cw_lds = reinterpret_cast<uint4*>(smem);  // 128 entries
-  uint4* pp = cw_lds + 128;                        // 2*Z ping-pong
-  uint4* stack = pp + 2 * Z;                       // Z*(DS-1)
-  u32* aes_lds =
-      reinterpret_cast<u32*>(stack + (size_t)Z * (DS > 1 ? DS - 1 : 0));
-  u32* red = aes_lds + (PRF == PRF_AES128 ? 1280 : 0);
-
-  // Stage codewords (512 ints) and AES tables into LDS.
-  for (int idx = t; idx < 128; idx += blockDim.x)
-    cw_lds[idx] =
-        reinterpret_cast<const uint4*>(keys + key_base + 4)[idx];
-  if constexpr (PRF == PRF_AES128) {
-    for (int idx = t; idx < 1280; idx += blockDim.x) aes_lds[idx] = aes_tabs[idx];
-  }
-  if (t == 0) {
-    const int* rp = keys + key_base + 516;
-    pp[0] = make_uint4((u32)rp[0], (u32)rp[1], (u32)rp[2], (u32)rp[3]);
-  }
-  __syncthreads();
-
-  // Phase 1: root -> Z frontier seeds (frontier position t has the
-  // first-consumed index bit as its MSB: t = bitrev(idx & (Z-1))).
-  uint4* a = pp;
-  uint4* b = pp + Z;
-  for (int l = 1; l <= zlog; ++l) {
-    const int i = depth - l;
-    if (t < (1 << l)) {
-      uint4 parent = a[t >> 1];
-      uint4 v = prf_full<PRF>(parent, (u32)(t & 1), aes_lds);
-      const int sel = (int)(parent.x & 1u);
-      b[t] = add128(v, cw_lds[sel * 64 + i * 2 + (t & 1)]);
-    }
-    __syncthreads();
-    uint4* tmp = a;
-    a = b;
-    b = tmp;
-  }
-  uint4 cur = a[t];
-  __syncthreads();
-
-  // Phase 2: initial descent to the first leaf-pair parent.
-  for (int d = 1; d <= DS - 1; ++d) {
-    uint4 c0, c1;
-    expand_pair<PRF>(cur, DS - d, cw_lds, aes_lds, c0, c1);
-    stack[(d - 1) * Z + t] = c1;
-    cur = c0;
-  }
-
-  u32 acc[16];
-#pragma unroll
-  for (int w = 0; w < 16; ++w) acc[w] = 0;
-
-  const long long pairs = 1LL << (DS - 1);
-  for (long long j = 0; j < pairs; ++j) {
-    u32 v0, v1;
-    expand_leaf_low<PRF>(cur, cw_lds, aes_lds, v0, v1);
-
-    const long long row = (j << (zlog + 1)) + ((long long)t << 1);
-    if constexpr (FUSED) {
-      const uint4* rows = reinterpret_cast<const uint4*>(table + row * 16);
-#pragma unroll
-      for (int q = 0; q < 4; ++q) {
-        uint4 r0 = rows[q];
-        uint4 r1 = rows[q + 4];
-        acc[4 * q + 0] += v0 * r0.x + v1 * r1.x;
-        acc[4 * q + 1] += v0 * r0.y + v1 * r1.y;
-        acc[4 * q + 2] += v0 * r0.z + v1 * r1.z;
-        acc[4 * q + 3] += v0 * r0.w + v1 * r1.w;
-      }
-    } else {
-      u32* orow = out + (u64)blockIdx.x * (u64)n + (u64)row;
-      orow[0] = v0;
-      orow[1] = v1;
-    }
-
-    if (j + 1 == pairs) break;
-    // Pop the deepest pending sibling and descend its bit-0 spine.
-    const int c = (int)(__ffsll((unsigned long long)(j + 1)) - 1);
-    const int dpop = DS - 1 - c;
-    cur = stack[(dpop - 1) * Z + t];
-    for (int d = dpop + 1; d <= DS - 1; ++d) {
-      uint4 c0, c1;
-      expand_pair<PRF>(cur, DS - d, cw_lds, aes_lds, c0, c1);
-      stack[(d - 1) * Z + t] = c1;
-      cur = c0;
-    }
-  }
-
-  if constexpr (FUSED) {
-    // Wave shfl reduction, then cross-wave sum through LDS.
-#pragma unroll
-    for (int w = 0; w < 16; ++w) {
-      u32 v = acc[w];
-#pragma unroll
-      for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
-      acc[w] = v;
-    }
-    const int lane = t & 63, wave = t >> 6, nwaves = (int)blockDim.x >> 6;
-    if (lane == 0) {
-#pragma unroll
-      for (int w = 0; w < 16; ++w) red[wave * 16 + w] = acc[w];
-    }
-    __syncthreads();
-    if (t < 16) {
-      u32 s = 0;
-      for (int wv = 0; wv < nwaves; ++wv) s += red[wv * 16 + t];
-      out[(u64)blockIdx.x * 16 + t] = s;
-    }
-  }
-}
-
-// ---------------------------------------------------------------------------
-// Naive oracle kernel: EvaluateFlat per (key, leaf), natural order.
-// ---------------------------------------------------------------------------
-template <int PRF>
-__global__ __launch_bounds__(256) void dpf_naive_kernel(const int* __restrict__ keys,
-                                 u32* __restrict__ out,
-                                 const u32* __restrict__ aes_tabs, int depth,
-                                 long long n) {
+__global__ __launch_bounds__(256) void dpf_naive_kernel(
+    const int* __restrict__ keys, u32* __restrict__ out,
+    const u32* __restrict__ aes_tabs, int depth, long long n) {
   extern __shared__ u32 smem[];
   uint4* cw_lds = reinterpret_cast<uint4*>(smem);
   u32* aes_lds = reinterpret_cast<u32*>(cw_lds + 128);
@@ -448,7 +699,8 @@ __global__ __launch_bounds__(256) void dpf_naive_kernel(const int* __restrict__ 
   for (int idx = t; idx < 128; idx += blockDim.x)
     cw_lds[idx] = reinterpret_cast<const uint4*>(keys + key_base + 4)[idx];
   if constexpr (PRF == PRF_AES128) {
-    for (int idx = t; idx < 1280; idx += blockDim.x) aes_lds[idx] = aes_tabs[idx];
+    for (int idx = t; idx < 1280; idx += blockDim.x)
+      aes_lds[idx] = aes_tabs[idx];
   }
   __syncthreads();
 
@@ -459,7 +711,7 @@ __global__ __launch_bounds__(256) void dpf_naive_kernel(const int* __restrict__ 
   long long rem = leaf;
   for (int i = depth - 1; i >= 0; --i) {
     const u32 bit = (u32)(rem & 1);
-    uint4 v = prf_full<PRF>(key, bit, aes_lds);
+    uint4 v = prf_naive<PRF>(aes_lds, key, bit);
     const int sel = (int)(key.x & 1u);
     key = add128(v, cw_lds[sel * 64 + i * 2 + (int)bit]);
     rem >>= 1;
@@ -473,11 +725,11 @@ __global__ __launch_bounds__(256) void dpf_naive_kernel(const int* __restrict__ 
 namespace {
 
 size_t fused_shmem_bytes(int Z, int DS, int prf) {
-  size_t bytes = 128 * 16;                       // codewords
-  bytes += (size_t)2 * Z * 16;                   // phase-1 ping-pong
-  bytes += (size_t)Z * (DS > 1 ? DS - 1 : 0) * 16;  // DFS sibling stack
-  if (prf == PRF_AES128) bytes += 1280 * 4;      // AES T-tables + sbox
-  bytes += (size_t)(Z / 64) * 16 * 4;            // cross-wave reduction
+  const int lds_levels = DS > 3 ? DS - 3 : 0;
+  size_t bytes = 128 * 16;                                  // codewords
+  bytes += (size_t)Z * (lds_levels > 2 ? lds_levels : 2) * 16;  // pp/stack
+  if (prf == PRF_AES128) bytes += AES_LDS_WORDS * 4;        // replicated te0
+  bytes += (size_t)(Z / 64) * 16 * 4;                       // reduction
   return bytes;
 }
 
@@ -487,6 +739,13 @@ void launch_eval_t(const int* keys, const u32* table, u32* out,
                    int zlog, hipStream_t stream) {
   const int Z = 1 << zlog;
   const int DS = depth - zlog;
+  // j-split: grow the grid to >= 1024 workgroups (4 per CU) so small
+  // batches still fill the chip; each extra split costs one targeted
+  // descent (DS-1 pair expansions) per workgroup — negligible against
+  // pairs/2^slog leaf pairs.
+  int slog = 0;
+  while ((batch << (slog + 1)) <= 2048 && slog + 2 <= DS - 1 && slog < 8)
+    ++slog;
   const size_t shmem = fused_shmem_bytes(Z, DS, PRF);
   auto kern = dpf_eval_kernel<PRF, FUSED>;
   if (shmem > 65536) {
@@ -494,8 +753,9 @@ void launch_eval_t(const int* keys, const u32* table, u32* out,
                                   hipFuncAttributeMaxDynamicSharedMemorySize,
                                   (int)shmem));
   }
-  hipLaunchKernelGGL(kern, dim3((unsigned)batch), dim3((unsigned)Z), shmem,
-                     stream, keys, table, out, aes_tabs, depth, zlog, n);
+  hipLaunchKernelGGL(kern, dim3((unsigned)(batch << slog)), dim3((unsigned)Z),
+                     shmem, stream, keys, table, out, aes_tabs, depth, zlog,
+                     slog, n);
   HIP_CHECK(hipGetLastError());
 }
 

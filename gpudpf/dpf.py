@@ -163,7 +163,8 @@ class DPF(object):
                 # rows are in leaf_perm order; gather back to natural order
                 out = out.index_select(1, self._perm_gpu)
             else:
-                out = torch.empty((b, self.ENTRY_SIZE), dtype=torch.int32, device=dev)
+                # zeroed: the kernel's j-split segments accumulate with atomics
+                out = torch.zeros((b, self.ENTRY_SIZE), dtype=torch.int32, device=dev)
                 _hip.eval_fused(
                     chunk.data_ptr(), self._table_gpu.data_ptr(), out.data_ptr(),
                     self._aes_ptr, b, n, depth, self._zlog, self.prf_method,
@@ -181,6 +182,7 @@ class DPF(object):
         b = keys_gpu.shape[0]
         dev = self._table_gpu.device
         stream = torch.cuda.current_stream(dev).cuda_stream
+        out_gpu.zero_()  # j-split segments accumulate with atomics
         _hip.eval_fused(
             keys_gpu.data_ptr(), self._table_gpu.data_ptr(), out_gpu.data_ptr(),
             self._aes_ptr, b, self.table_num_entries, self._depth, self._zlog,
