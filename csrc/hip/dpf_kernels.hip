@@ -240,122 +240,79 @@ struct AesLds {
   __device__ __forceinline__ u32 sbox(u32 x) const { return (te0(x) >> 16) & 0xff; }
 };
 
-__device__ __forceinline__ void aes_expand_rk(uint4 seed, const AesLds& T,
-                                              u32 rk[44]) {
-  rk[0] = __builtin_bswap32(seed.x);
-  rk[1] = __builtin_bswap32(seed.y);
-  rk[2] = __builtin_bswap32(seed.z);
-  rk[3] = __builtin_bswap32(seed.w);
+// Both children (pos 0 and pos 1) with an ON-THE-FLY key schedule fused
+// into the interleaved cipher rounds: the key-expansion sbox chain (4
+// dependent LDS lookups per round) overlaps the 32 independent cipher
+// lookups of the same round instead of standing alone as a 40-deep serial
+// latency chain (PMC: SQ_WAIT_ANY was 31.5% of wave cycles with the
+// standalone schedule).  Also drops the rk[44] register array.
+template <bool LOW>
+__device__ __forceinline__ void aes_cipher_pair(uint4 seed, const AesLds& T,
+                                                uint4& ra, uint4& rb) {
+  u32 k0 = __builtin_bswap32(seed.x), k1 = __builtin_bswap32(seed.y),
+      k2 = __builtin_bswap32(seed.z), k3 = __builtin_bswap32(seed.w);
+  u32 s0 = k0, s1 = k1, s2 = k2, s3 = k3;
+  u32 u0 = (1u << 24) ^ k0, u1 = k1, u2 = k2, u3 = k3;
   u32 rcon = 0x01u;
 #pragma unroll
-  for (int r = 1; r <= 10; ++r) {
-    u32 w = rk[4 * r - 1];
-    w = (w << 8) | (w >> 24);
-    w = (T.sbox((w >> 24) & 0xff) << 24) | (T.sbox((w >> 16) & 0xff) << 16) |
-        (T.sbox((w >> 8) & 0xff) << 8) | T.sbox(w & 0xff);
-    w ^= (rcon << 24);
+  for (int r = 1; r < 10; ++r) {
+    // next round key (k0..k3 become rk[4r..4r+3])
+    u32 w_ = (k3 << 8) | (k3 >> 24);
+    w_ = (T.sbox((w_ >> 24) & 0xff) << 24) | (T.sbox((w_ >> 16) & 0xff) << 16) |
+         (T.sbox((w_ >> 8) & 0xff) << 8) | T.sbox(w_ & 0xff);
+    w_ ^= (rcon << 24);
     rcon = (rcon << 1) ^ ((rcon & 0x80u) ? 0x11bu : 0u);
     rcon &= 0xffu;
-    rk[4 * r] = rk[4 * r - 4] ^ w;
-    rk[4 * r + 1] = rk[4 * r - 3] ^ rk[4 * r];
-    rk[4 * r + 2] = rk[4 * r - 2] ^ rk[4 * r + 1];
-    rk[4 * r + 3] = rk[4 * r - 1] ^ rk[4 * r + 2];
-  }
-}
-
-// LOW=true computes only the low 32 bits of the result (4 final-round
-// lookups instead of 16).
-template <bool LOW>
-__device__ __forceinline__ uint4 aes_cipher_rep(const u32 rk[44],
-                                                const AesLds& T, u32 pos) {
-  u32 s0 = (pos << 24) ^ rk[0], s1 = rk[1], s2 = rk[2], s3 = rk[3];
-#pragma unroll
-  for (int r = 1; r < 10; ++r) {
+    k0 ^= w_; k1 ^= k0; k2 ^= k1; k3 ^= k2;
+    // cipher round r for both children
     u32 n0 = T.te0(s0 >> 24) ^ rotr8(T.te0((s1 >> 16) & 0xff)) ^
-             rotr16(T.te0((s2 >> 8) & 0xff)) ^ rotr24(T.te0(s3 & 0xff)) ^
-             rk[4 * r];
-    u32 n1 = T.te0(s1 >> 24) ^ rotr8(T.te0((s2 >> 16) & 0xff)) ^
-             rotr16(T.te0((s3 >> 8) & 0xff)) ^ rotr24(T.te0(s0 & 0xff)) ^
-             rk[4 * r + 1];
-    u32 n2 = T.te0(s2 >> 24) ^ rotr8(T.te0((s3 >> 16) & 0xff)) ^
-             rotr16(T.te0((s0 >> 8) & 0xff)) ^ rotr24(T.te0(s1 & 0xff)) ^
-             rk[4 * r + 2];
-    u32 n3 = T.te0(s3 >> 24) ^ rotr8(T.te0((s0 >> 16) & 0xff)) ^
-             rotr16(T.te0((s1 >> 8) & 0xff)) ^ rotr24(T.te0(s2 & 0xff)) ^
-             rk[4 * r + 3];
-    s0 = n0; s1 = n1; s2 = n2; s3 = n3;
-  }
-  u32 o0 = ((T.sbox(s0 >> 24) << 24) | (T.sbox((s1 >> 16) & 0xff) << 16) |
-            (T.sbox((s2 >> 8) & 0xff) << 8) | T.sbox(s3 & 0xff)) ^ rk[40];
-  if constexpr (LOW) return make_uint4(__builtin_bswap32(o0), 0, 0, 0);
-  u32 o1 = ((T.sbox(s1 >> 24) << 24) | (T.sbox((s2 >> 16) & 0xff) << 16) |
-            (T.sbox((s3 >> 8) & 0xff) << 8) | T.sbox(s0 & 0xff)) ^ rk[41];
-  u32 o2 = ((T.sbox(s2 >> 24) << 24) | (T.sbox((s3 >> 16) & 0xff) << 16) |
-            (T.sbox((s0 >> 8) & 0xff) << 8) | T.sbox(s1 & 0xff)) ^ rk[42];
-  u32 o3 = ((T.sbox(s3 >> 24) << 24) | (T.sbox((s0 >> 16) & 0xff) << 16) |
-            (T.sbox((s1 >> 8) & 0xff) << 8) | T.sbox(s2 & 0xff)) ^ rk[43];
-  return make_uint4(__builtin_bswap32(o0), __builtin_bswap32(o1),
-                    __builtin_bswap32(o2), __builtin_bswap32(o3));
-}
-
-// Both children (pos 0 and pos 1) under one key schedule, rounds
-// interleaved for 2x ILP on both the LDS and VALU pipes.
-template <bool LOW>
-__device__ __forceinline__ void aes_cipher_pair(const u32 rk[44],
-                                                const AesLds& T, uint4& ra,
-                                                uint4& rb) {
-  u32 s0 = rk[0], s1 = rk[1], s2 = rk[2], s3 = rk[3];
-  u32 u0 = (1u << 24) ^ rk[0], u1 = rk[1], u2 = rk[2], u3 = rk[3];
-#pragma unroll
-  for (int r = 1; r < 10; ++r) {
-    u32 n0 = T.te0(s0 >> 24) ^ rotr8(T.te0((s1 >> 16) & 0xff)) ^
-             rotr16(T.te0((s2 >> 8) & 0xff)) ^ rotr24(T.te0(s3 & 0xff)) ^
-             rk[4 * r];
+             rotr16(T.te0((s2 >> 8) & 0xff)) ^ rotr24(T.te0(s3 & 0xff)) ^ k0;
     u32 m0 = T.te0(u0 >> 24) ^ rotr8(T.te0((u1 >> 16) & 0xff)) ^
-             rotr16(T.te0((u2 >> 8) & 0xff)) ^ rotr24(T.te0(u3 & 0xff)) ^
-             rk[4 * r];
+             rotr16(T.te0((u2 >> 8) & 0xff)) ^ rotr24(T.te0(u3 & 0xff)) ^ k0;
     u32 n1 = T.te0(s1 >> 24) ^ rotr8(T.te0((s2 >> 16) & 0xff)) ^
-             rotr16(T.te0((s3 >> 8) & 0xff)) ^ rotr24(T.te0(s0 & 0xff)) ^
-             rk[4 * r + 1];
+             rotr16(T.te0((s3 >> 8) & 0xff)) ^ rotr24(T.te0(s0 & 0xff)) ^ k1;
     u32 m1 = T.te0(u1 >> 24) ^ rotr8(T.te0((u2 >> 16) & 0xff)) ^
-             rotr16(T.te0((u3 >> 8) & 0xff)) ^ rotr24(T.te0(u0 & 0xff)) ^
-             rk[4 * r + 1];
+             rotr16(T.te0((u3 >> 8) & 0xff)) ^ rotr24(T.te0(u0 & 0xff)) ^ k1;
     u32 n2 = T.te0(s2 >> 24) ^ rotr8(T.te0((s3 >> 16) & 0xff)) ^
-             rotr16(T.te0((s0 >> 8) & 0xff)) ^ rotr24(T.te0(s1 & 0xff)) ^
-             rk[4 * r + 2];
+             rotr16(T.te0((s0 >> 8) & 0xff)) ^ rotr24(T.te0(s1 & 0xff)) ^ k2;
     u32 m2 = T.te0(u2 >> 24) ^ rotr8(T.te0((u3 >> 16) & 0xff)) ^
-             rotr16(T.te0((u0 >> 8) & 0xff)) ^ rotr24(T.te0(u1 & 0xff)) ^
-             rk[4 * r + 2];
+             rotr16(T.te0((u0 >> 8) & 0xff)) ^ rotr24(T.te0(u1 & 0xff)) ^ k2;
     u32 n3 = T.te0(s3 >> 24) ^ rotr8(T.te0((s0 >> 16) & 0xff)) ^
-             rotr16(T.te0((s1 >> 8) & 0xff)) ^ rotr24(T.te0(s2 & 0xff)) ^
-             rk[4 * r + 3];
+             rotr16(T.te0((s1 >> 8) & 0xff)) ^ rotr24(T.te0(s2 & 0xff)) ^ k3;
     u32 m3 = T.te0(u3 >> 24) ^ rotr8(T.te0((u0 >> 16) & 0xff)) ^
-             rotr16(T.te0((u1 >> 8) & 0xff)) ^ rotr24(T.te0(u2 & 0xff)) ^
-             rk[4 * r + 3];
+             rotr16(T.te0((u1 >> 8) & 0xff)) ^ rotr24(T.te0(u2 & 0xff)) ^ k3;
     s0 = n0; s1 = n1; s2 = n2; s3 = n3;
     u0 = m0; u1 = m1; u2 = m2; u3 = m3;
   }
+  // round-10 key
+  {
+    u32 w_ = (k3 << 8) | (k3 >> 24);
+    w_ = (T.sbox((w_ >> 24) & 0xff) << 24) | (T.sbox((w_ >> 16) & 0xff) << 16) |
+         (T.sbox((w_ >> 8) & 0xff) << 8) | T.sbox(w_ & 0xff);
+    w_ ^= (rcon << 24);
+    k0 ^= w_; k1 ^= k0; k2 ^= k1; k3 ^= k2;
+  }
   u32 o0 = ((T.sbox(s0 >> 24) << 24) | (T.sbox((s1 >> 16) & 0xff) << 16) |
-            (T.sbox((s2 >> 8) & 0xff) << 8) | T.sbox(s3 & 0xff)) ^ rk[40];
+            (T.sbox((s2 >> 8) & 0xff) << 8) | T.sbox(s3 & 0xff)) ^ k0;
   u32 p0 = ((T.sbox(u0 >> 24) << 24) | (T.sbox((u1 >> 16) & 0xff) << 16) |
-            (T.sbox((u2 >> 8) & 0xff) << 8) | T.sbox(u3 & 0xff)) ^ rk[40];
+            (T.sbox((u2 >> 8) & 0xff) << 8) | T.sbox(u3 & 0xff)) ^ k0;
   if constexpr (LOW) {
     ra = make_uint4(__builtin_bswap32(o0), 0, 0, 0);
     rb = make_uint4(__builtin_bswap32(p0), 0, 0, 0);
     return;
   }
   u32 o1 = ((T.sbox(s1 >> 24) << 24) | (T.sbox((s2 >> 16) & 0xff) << 16) |
-            (T.sbox((s3 >> 8) & 0xff) << 8) | T.sbox(s0 & 0xff)) ^ rk[41];
+            (T.sbox((s3 >> 8) & 0xff) << 8) | T.sbox(s0 & 0xff)) ^ k1;
   u32 p1 = ((T.sbox(u1 >> 24) << 24) | (T.sbox((u2 >> 16) & 0xff) << 16) |
-            (T.sbox((u3 >> 8) & 0xff) << 8) | T.sbox(u0 & 0xff)) ^ rk[41];
+            (T.sbox((u3 >> 8) & 0xff) << 8) | T.sbox(u0 & 0xff)) ^ k1;
   u32 o2 = ((T.sbox(s2 >> 24) << 24) | (T.sbox((s3 >> 16) & 0xff) << 16) |
-            (T.sbox((s0 >> 8) & 0xff) << 8) | T.sbox(s1 & 0xff)) ^ rk[42];
+            (T.sbox((s0 >> 8) & 0xff) << 8) | T.sbox(s1 & 0xff)) ^ k2;
   u32 p2 = ((T.sbox(u2 >> 24) << 24) | (T.sbox((u3 >> 16) & 0xff) << 16) |
-            (T.sbox((u0 >> 8) & 0xff) << 8) | T.sbox(u1 & 0xff)) ^ rk[42];
+            (T.sbox((u0 >> 8) & 0xff) << 8) | T.sbox(u1 & 0xff)) ^ k2;
   u32 o3 = ((T.sbox(s3 >> 24) << 24) | (T.sbox((s0 >> 16) & 0xff) << 16) |
-            (T.sbox((s1 >> 8) & 0xff) << 8) | T.sbox(s2 & 0xff)) ^ rk[43];
+            (T.sbox((s1 >> 8) & 0xff) << 8) | T.sbox(s2 & 0xff)) ^ k3;
   u32 p3 = ((T.sbox(u3 >> 24) << 24) | (T.sbox((u0 >> 16) & 0xff) << 16) |
-            (T.sbox((u1 >> 8) & 0xff) << 8) | T.sbox(u2 & 0xff)) ^ rk[43];
+            (T.sbox((u1 >> 8) & 0xff) << 8) | T.sbox(u2 & 0xff)) ^ k3;
   ra = make_uint4(__builtin_bswap32(o0), __builtin_bswap32(o1),
                   __builtin_bswap32(o2), __builtin_bswap32(o3));
   rb = make_uint4(__builtin_bswap32(p0), __builtin_bswap32(p1),
@@ -372,9 +329,9 @@ __device__ __forceinline__ uint4 prf_full(uint4 seed, u32 pos,
   if constexpr (PRF == PRF_SALSA20) return salsa12_core(seed, pos);
   if constexpr (PRF == PRF_CHACHA20) return chacha12_core(seed, pos);
   if constexpr (PRF == PRF_AES128) {
-    u32 rk[44];
-    aes_expand_rk(seed, T, rk);
-    return aes_cipher_rep<false>(rk, T, pos);
+    uint4 a, b;
+    aes_cipher_pair<false>(seed, T, a, b);
+    return pos ? b : a;
   }
 }
 
@@ -382,9 +339,7 @@ template <int PRF>
 __device__ __forceinline__ void prf_pair(uint4 seed, const AesLds& T,
                                          uint4& r0, uint4& r1) {
   if constexpr (PRF == PRF_AES128) {
-    u32 rk[44];
-    aes_expand_rk(seed, T, rk);
-    aes_cipher_pair<false>(rk, T, r0, r1);
+    aes_cipher_pair<false>(seed, T, r0, r1);
   } else if constexpr (PRF == PRF_SALSA20) {
     salsa12_pair_core(seed, r0, r1);
   } else if constexpr (PRF == PRF_CHACHA20) {
@@ -412,10 +367,8 @@ __device__ __forceinline__ void prf_pair_low(uint4 seed, const AesLds& T,
     r0 = a.x;
     r1 = b.x;
   } else {
-    u32 rk[44];
-    aes_expand_rk(seed, T, rk);
     uint4 a, b;
-    aes_cipher_pair<true>(rk, T, a, b);
+    aes_cipher_pair<true>(seed, T, a, b);
     r0 = a.x;
     r1 = b.x;
   }
