@@ -1,0 +1,55 @@
+"""Headline benchmark sweep (parity with the reference's benchmark.py:
+sweep table sizes x PRFs, print DPFs/sec per config).
+
+Usage: python benchmark.py [--sizes 16384,65536,262144,1048576]
+                           [--prfs AES128,SALSA20,CHACHA20] [--batch 512]
+"""
+
+import argparse
+import time
+
+import torch
+
+from gpudpf import DPF
+
+PRF_IDS = {
+    "DUMMY": DPF.PRF_DUMMY,
+    "SALSA20": DPF.PRF_SALSA20,
+    "CHACHA20": DPF.PRF_CHACHA20,
+    "AES128": DPF.PRF_AES128,
+}
+
+
+def test_gpu_dpf_perf(N, batch=512, entrysize=16, prf=DPF.PRF_AES128, reps=10):
+    dpf = DPF(prf=prf)
+    k1, _ = dpf.gen(1, N)
+    keys = torch.stack([k1] * batch)
+    table = torch.randint(-(2**31), 2**31 - 1, (N, entrysize), dtype=torch.int64).to(
+        torch.int32
+    )
+    dpf.eval_init(table)
+    dpf.eval_gpu(keys)  # warmup
+    torch.cuda.synchronize()
+    tstart = time.time()
+    for _ in range(reps):
+        dpf.eval_gpu(keys)
+    torch.cuda.synchronize()
+    elapsed = time.time() - tstart
+    dpfs_per_sec = batch * reps / elapsed
+    keysize = int(k1.numel()) * 4
+    print(
+        "%s Key Size: %d bytes, Perf: %d dpfs/sec"
+        % (dpf, keysize, dpfs_per_sec)
+    )
+    return dpfs_per_sec
+
+
+if __name__ == "__main__":
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--sizes", default="16384,65536,262144,1048576")
+    ap.add_argument("--prfs", default="AES128,SALSA20,CHACHA20")
+    ap.add_argument("--batch", type=int, default=512)
+    args = ap.parse_args()
+    for prf_name in args.prfs.split(","):
+        for n in (int(s) for s in args.sizes.split(",")):
+            test_gpu_dpf_perf(N=n, batch=args.batch, prf=PRF_IDS[prf_name])

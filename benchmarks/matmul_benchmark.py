@@ -1,0 +1,54 @@
+"""GEMM128 benchmark (the reference's matmul_benchmark.cu analog):
+throughput of the exact mod-2^128 GEMM at PIR-shaped sizes
+(M=batch, N=entry words, K=table entries), dict-line output."""
+
+import argparse
+import time
+
+import torch
+
+from gpudpf import ops
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--m", type=int, default=512)
+    ap.add_argument("--n", type=int, default=16)
+    ap.add_argument("--k", type=int, default=65536)
+    ap.add_argument("--reps", type=int, default=5)
+    ap.add_argument("--check", action="store_true")
+    a = ap.parse_args()
+
+    torch.manual_seed(0)
+    A = torch.randint(-(2**31), 2**31 - 1, (a.m, a.k, 4), dtype=torch.int64).to(
+        torch.int32
+    )
+    Bt = torch.randint(-(2**31), 2**31 - 1, (a.n, a.k, 4), dtype=torch.int64).to(
+        torch.int32
+    )
+    if a.check:
+        got = ops.gemm128(A, Bt).cpu()
+        want = ops.gemm128_cpu(A, Bt)
+        assert torch.equal(got, want), "gemm128 check failed"
+        print("check OK")
+
+    dev = torch.device("cuda:0")
+    A_g, B_g = A.to(dev), Bt.to(dev)
+    ops.gemm128(A_g, B_g)
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(a.reps):
+        ops.gemm128(A_g, B_g)
+    torch.cuda.synchronize()
+    dt = (time.time() - t0) / a.reps
+    macs = a.m * a.n * a.k
+    print({
+        "kernel": "gemm128",
+        "m": a.m, "n": a.n, "k": a.k, "reps": a.reps,
+        "time_ms": round(dt * 1e3, 3),
+        "gmacs128_per_sec": round(macs / dt / 1e9, 2),
+    })
+
+
+if __name__ == "__main__":
+    main()

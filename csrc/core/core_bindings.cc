@@ -148,6 +148,31 @@ KeyArr shard_subkey(const KeyArr& key, int prf_method, u64 rank, u64 world) {
   return key_to_array(sub);
 }
 
+// Exact u128 GEMM CPU reference: a [M,K,4] int32 (u128 limbs LE),
+// bt [N,K,4] -> c [M,N,4].
+py::array_t<std::int32_t> gemm128_cpu(
+    py::array_t<std::int32_t, py::array::c_style | py::array::forcecast> a,
+    py::array_t<std::int32_t, py::array::c_style | py::array::forcecast> bt) {
+  if (a.ndim() != 3 || bt.ndim() != 3 || a.shape(2) != 4 || bt.shape(2) != 4 ||
+      a.shape(1) != bt.shape(1))
+    throw std::invalid_argument("shapes must be [M,K,4] and [N,K,4]");
+  const py::ssize_t M = a.shape(0), N = bt.shape(0), K = a.shape(1);
+  py::array_t<std::int32_t> c({M, N, (py::ssize_t)4});
+  const u128* ap = reinterpret_cast<const u128*>(a.data());
+  const u128* bp = reinterpret_cast<const u128*>(bt.data());
+  u128* cp = reinterpret_cast<u128*>(c.mutable_data());
+  {
+    py::gil_scoped_release nogil;
+    for (py::ssize_t m = 0; m < M; ++m)
+      for (py::ssize_t n = 0; n < N; ++n) {
+        u128 acc = 0;
+        for (py::ssize_t k = 0; k < K; ++k) acc += ap[m * K + k] * bp[n * K + k];
+        cp[m * N + n] = acc;
+      }
+  }
+  return c;
+}
+
 // AES GPU tables (5 x 256 u32: te0..te3, sbox) for upload to the device.
 py::array_t<std::int32_t> aes_gpu_tables() {
   py::array_t<std::int32_t> out(5 * 256);
@@ -174,6 +199,7 @@ PYBIND11_MODULE(_core, m) {
   m.def("aes_block", &aes_block);
   m.def("shard_subkey", &shard_subkey);
   m.def("aes_gpu_tables", &aes_gpu_tables);
+  m.def("gemm128_cpu", &gemm128_cpu);
   m.attr("KEY_INTS") = py::int_(kKeyInts);
   m.attr("ENTRY_WORDS") = py::int_(kEntryWords);
   m.attr("PRF_DUMMY") = py::int_((int)PRF_DUMMY);

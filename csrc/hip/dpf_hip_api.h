@@ -25,4 +25,12 @@ void launch_naive(std::uintptr_t keys, std::uintptr_t out,
                   std::uintptr_t aes_tabs, int batch, long long n, int depth,
                   int prf, std::uintptr_t stream);
 
+// Exact u128 GEMM (research harness; gemm128.hip).  a: [M,K] u128 (as
+// 4xint32 limbs), bt: [N,K] u128, c: [M,N] u128, partials: device scratch
+// of gemm128_ksplit(M,N,K)*M*N u128.
+int gemm128_ksplit(long long M, long long N, long long K);
+void launch_gemm128(std::uintptr_t a, std::uintptr_t bt, std::uintptr_t c,
+                    std::uintptr_t partials, long long M, long long N,
+                    long long K, std::uintptr_t stream);
+
 }  // namespace gpudpf_hip

@@ -71,6 +71,11 @@ PYBIND11_MODULE(_hip, m) {
         py::arg("out"), py::arg("aes_tabs"), py::arg("batch"), py::arg("n"),
         py::arg("depth"), py::arg("prf"), py::arg("stream"),
         py::call_guard<py::gil_scoped_release>());
+  m.def("gemm128", &gpudpf_hip::launch_gemm128, py::arg("a"), py::arg("bt"),
+        py::arg("c"), py::arg("partials"), py::arg("m"), py::arg("n"),
+        py::arg("k"), py::arg("stream"),
+        py::call_guard<py::gil_scoped_release>());
+  m.def("gemm128_ksplit", &gpudpf_hip::gemm128_ksplit);
   m.def("ensure_aes_tables", &ensure_aes_tables, py::arg("device"));
   m.def("device_count", &device_count);
 }
