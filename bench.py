@@ -79,7 +79,7 @@ def main():
     # Synthetic workload: random-init table, random target indices.
     torch.manual_seed(1000 + rank)
     table = torch.randint(-(2**31), 2**31 - 1, (n, args.entry_size),
-                          dtype=torch.int64).to(torch.int32)
+                          dtype=torch.int32)
     keys = []
     g = torch.Generator().manual_seed(rank)
     for i in range(args.batch):

@@ -9,6 +9,12 @@ Usage: python benchmarks/cpu_benchmark.py [--n 16384] [--batch 512]
 Thread sweep: bash benchmarks/cpu_thread_sweep.sh
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 import time
 

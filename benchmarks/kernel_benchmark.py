@@ -15,6 +15,12 @@ Strategies:
   naive  - O(n log n) per-leaf oracle kernel
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 import time
 

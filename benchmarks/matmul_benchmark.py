@@ -2,6 +2,12 @@
 throughput of the exact mod-2^128 GEMM at PIR-shaped sizes
 (M=batch, N=entry words, K=table entries), dict-line output."""
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import argparse
 import time
 

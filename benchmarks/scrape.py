@@ -5,6 +5,12 @@ kernel_benchmark.py; parse (ast.literal_eval, not eval) and emit CSV.
 Usage: python benchmarks/scrape.py <dir-or-files...> [> sweep.csv]
 """
 
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+
 import ast
 import csv
 import os
