@@ -16,6 +16,41 @@ import torch
 from gpudpf import ops
 
 
+def bench_u32_mfma(a):
+    import numpy as np
+
+    torch.manual_seed(0)
+    A = torch.randint(-(2**31), 2**31 - 1, (a.m, a.k), dtype=torch.int64).to(
+        torch.int32
+    )
+    B = torch.randint(-(2**31), 2**31 - 1, (a.k, a.n), dtype=torch.int64).to(
+        torch.int32
+    )
+    if a.check:
+        got = ops.pir_matmul_u32(A, B).cpu().numpy()
+        want = (A.numpy().astype(np.int64) @ B.numpy().astype(np.int64)).astype(
+            np.uint32
+        ).astype(np.int32)
+        assert np.array_equal(got, want), "u32 mfma check failed"
+        print("check OK")
+    dev = torch.device("cuda:0")
+    A_g, B_g = A.to(dev), B.to(dev)
+    ops.pir_matmul_u32(A_g, B_g)
+    torch.cuda.synchronize()
+    t0 = time.time()
+    for _ in range(a.reps):
+        ops.pir_matmul_u32(A_g, B_g)
+    torch.cuda.synchronize()
+    dt = (time.time() - t0) / a.reps
+    macs = a.m * a.n * a.k
+    print({
+        "kernel": "gemm_u32_mfma",
+        "m": a.m, "n": a.n, "k": a.k, "reps": a.reps,
+        "time_ms": round(dt * 1e3, 3),
+        "gmacs32_per_sec": round(macs / dt / 1e9, 2),
+    })
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--m", type=int, default=512)
@@ -23,7 +58,11 @@ def main():
     ap.add_argument("--k", type=int, default=65536)
     ap.add_argument("--reps", type=int, default=5)
     ap.add_argument("--check", action="store_true")
+    ap.add_argument("--kernel", default="u128", choices=["u128", "u32mfma"])
     a = ap.parse_args()
+    if a.kernel == "u32mfma":
+        bench_u32_mfma(a)
+        return
 
     torch.manual_seed(0)
     A = torch.randint(-(2**31), 2**31 - 1, (a.m, a.k, 4), dtype=torch.int64).to(

@@ -33,4 +33,13 @@ void launch_gemm128(std::uintptr_t a, std::uintptr_t bt, std::uintptr_t c,
                     std::uintptr_t partials, long long M, long long N,
                     long long K, std::uintptr_t stream);
 
+// MFMA mod-2^32 GEMM (gemm_u32.hip): digit decomposition + i8 matrix
+// cores.  da/dbt: [4][M][K] / [4][N][K] int8 digit planes; c: [M][N] u32
+// (zeroed by caller; K-split partials combine with wrapping atomicAdd).
+void launch_digits(std::uintptr_t in, std::uintptr_t out, long long count,
+                   std::uintptr_t stream);
+void launch_gemm_u32_mfma(std::uintptr_t da, std::uintptr_t dbt,
+                          std::uintptr_t c, long long M, long long N,
+                          long long K, std::uintptr_t stream);
+
 }  // namespace gpudpf_hip

@@ -76,6 +76,12 @@ PYBIND11_MODULE(_hip, m) {
         py::arg("k"), py::arg("stream"),
         py::call_guard<py::gil_scoped_release>());
   m.def("gemm128_ksplit", &gpudpf_hip::gemm128_ksplit);
+  m.def("digits", &gpudpf_hip::launch_digits, py::arg("inp"), py::arg("out"),
+        py::arg("count"), py::arg("stream"),
+        py::call_guard<py::gil_scoped_release>());
+  m.def("gemm_u32_mfma", &gpudpf_hip::launch_gemm_u32_mfma, py::arg("da"),
+        py::arg("dbt"), py::arg("c"), py::arg("m"), py::arg("n"), py::arg("k"),
+        py::arg("stream"), py::call_guard<py::gil_scoped_release>());
   m.def("ensure_aes_tables", &ensure_aes_tables, py::arg("device"));
   m.def("device_count", &device_count);
 }

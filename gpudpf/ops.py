@@ -41,3 +41,40 @@ def gemm128_cpu(a, bt):
     return torch.from_numpy(
         np.asarray(_core.gemm128_cpu(a.numpy(), bt.numpy()))
     )
+
+
+def _digit_planes(x_u32_gpu):
+    """[..., count] int32 tensor on GPU -> [4, count] int8 signed base-256
+    digit planes (flattened over the input shape)."""
+    count = x_u32_gpu.numel()
+    out = torch.empty((4, count), dtype=torch.int8, device=x_u32_gpu.device)
+    stream = torch.cuda.current_stream(x_u32_gpu.device).cuda_stream
+    _hip.digits(x_u32_gpu.data_ptr(), out.data_ptr(), count, stream)
+    return out
+
+
+def pir_matmul_u32(shares, table):
+    """C = shares @ table mod 2^32 on the MFMA matrix cores.
+
+    shares: [M, K] int32 (e.g. one-hot DPF shares), table: [K, N] int32.
+    Each u32 operand is decomposed into 4 signed base-256 digit planes and
+    the product assembled from 10 i8-MFMA accumulator chains (see
+    csrc/hip/gemm_u32.hip).  Exact mod 2^32.
+    """
+    assert shares.dtype == torch.int32 and table.dtype == torch.int32
+    M, K = shares.shape
+    K2, N = table.shape
+    assert K == K2
+    dev = torch.device("cuda:0") if shares.device.type == "cpu" else shares.device
+    Mp, Np, Kp = -(-M // 64) * 64, -(-N // 16) * 16, -(-K // 64) * 64
+    a = torch.zeros((Mp, Kp), dtype=torch.int32, device=dev)
+    a[:M, :K] = shares.to(dev)
+    bt = torch.zeros((Np, Kp), dtype=torch.int32, device=dev)
+    bt[:N, :K] = table.to(dev).t()
+    da = _digit_planes(a)
+    dbt = _digit_planes(bt)
+    c = torch.zeros((Mp, Np), dtype=torch.int32, device=dev)
+    stream = torch.cuda.current_stream(dev).cuda_stream
+    _hip.gemm_u32_mfma(da.data_ptr(), dbt.data_ptr(), c.data_ptr(), Mp, Np,
+                       Kp, stream)
+    return c[:M, :N]
