@@ -130,11 +130,18 @@ class DPF(object):
         depth = int(kt[0, 0].item())
         return kt.contiguous(), n, depth
 
-    def eval_gpu(self, keys, one_hot_only=False):
+    def eval_gpu(self, keys, one_hot_only=False, strategy="fused"):
         """Evaluate a batch of keys against the initialized table on the
         GPU.  Returns [batch, e] int32 secret shares (CPU tensor), or the
         raw [batch, n] one-hot shares if one_hot_only (a capability the
-        reference lists as a TODO, dpf.py:30)."""
+        reference lists as a TODO, dpf.py:30).
+
+        strategy: "fused" (production: expansion fused with the table MAC)
+        or "two_stage" (expand one-hot shares, then multiply against the
+        table on the MFMA matrix cores — the runtime strategy selection the
+        reference leaves as a TODO, dpf.py:26)."""
+        if strategy == "two_stage" and not one_hot_only:
+            return self._eval_gpu_two_stage(keys)
         if self._table_gpu is None:
             raise Exception("Must call `eval_init` before `eval_gpu`")
         if not _HAS_HIP:
@@ -174,6 +181,23 @@ class DPF(object):
             results.append(out)
         res = torch.cat(results) if len(results) > 1 else results[0]
         return res.cpu()
+
+    def _eval_gpu_two_stage(self, keys):
+        """Expand one-hot shares (permuted rows), then reduce against the
+        permuted table with the MFMA mod-2^32 GEMM — no gather needed
+        because both sides share the leaf_perm row order."""
+        from gpudpf import ops
+
+        kt, n, depth = self._keys_tensor(keys)
+        dev = self._table_gpu.device
+        stream = torch.cuda.current_stream(dev).cuda_stream
+        keys_gpu = kt.to(dev, non_blocking=True).contiguous()
+        b = kt.shape[0]
+        shares = torch.empty((b, n), dtype=torch.int32, device=dev)
+        _hip.eval_expand(keys_gpu.data_ptr(), shares.data_ptr(), self._aes_ptr,
+                         b, n, depth, self._zlog, self.prf_method, stream)
+        out = ops.pir_matmul_u32(shares, self._table_gpu)
+        return out[:, : self.table_effective_entry_size].cpu()
 
     def eval_gpu_into(self, keys_gpu, out_gpu):
         """Zero-copy serving path: keys already on device as [b,524] int32,

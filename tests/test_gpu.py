@@ -162,3 +162,20 @@ def test_gpu_perf_smoke():
     torch.cuda.synchronize()
     dt = time.time() - t0
     print("AES128 n=%d: %.0f dpfs/sec" % (n, batch * reps / dt))
+
+
+def test_gpu_two_stage_strategy_matches_fused():
+    n = 16384
+    for prf in [DPF.PRF_SALSA20, DPF.PRF_AES128]:
+        dpf = DPF(prf=prf)
+        alpha = 777
+        k1, k2 = dpf.gen(alpha, n)
+        table = torch.randint(-(2**31), 2**31 - 1, (n, 16), dtype=torch.int64).to(
+            torch.int32
+        )
+        dpf.eval_init(table)
+        fused = dpf.eval_gpu([k1, k2])
+        two = dpf.eval_gpu([k1, k2], strategy="two_stage")
+        assert torch.equal(fused, two), prf
+        rec = (two[0].to(torch.int64) - two[1].to(torch.int64)).to(torch.int32)
+        assert torch.equal(rec, table[alpha])
