@@ -28,7 +28,11 @@ def test_gpu_dpf_perf(N, batch=512, entrysize=16, prf=DPF.PRF_AES128, reps=10):
         torch.int32
     )
     dpf.eval_init(table)
-    dpf.eval_gpu(keys)  # warmup
+    # warm up until clocks ramp (the GPU idles at low clock during CPU
+    # keygen; a single warmup step under-reports small-n throughput)
+    tw = time.time()
+    while time.time() - tw < 0.5:
+        dpf.eval_gpu(keys)
     torch.cuda.synchronize()
     tstart = time.time()
     for _ in range(reps):

@@ -93,7 +93,11 @@ def run(strategy, prf_name, n, batch, entry_size, reps, check=False):
             got = launch(1, keys_gpu[:1])[0].cpu()
             assert torch.equal(got, want_shares), "naive check failed"
 
-    torch.cuda.synchronize()
+    # clock-ramp warmup
+    tw = time.time()
+    while time.time() - tw < 0.4:
+        launch(batch, keys_gpu)
+        torch.cuda.synchronize()
     t0 = time.time()
     for _ in range(reps):
         launch(batch, keys_gpu)
