@@ -173,6 +173,54 @@ py::array_t<std::int32_t> gemm128_cpu(
   return c;
 }
 
+// sqrt(n) grid DPF: gen -> two dicts of numpy arrays; full-domain eval.
+py::tuple grid_gen(u64 alpha, u64 n_keys, u64 n_codewords, py::bytes seed,
+                   int prf_method) {
+  std::string sd = seed;
+  KeyRng rng(reinterpret_cast<const unsigned char*>(sd.data()), sd.size());
+  GridDpfKey k0, k1;
+  grid_dpf_gen(alpha, /*beta=*/1, n_keys, n_codewords, prf_method, rng, k0, k1);
+  auto pack = [](const GridDpfKey& k) {
+    py::dict d;
+    py::array_t<std::int32_t> seeds({(py::ssize_t)k.n_keys, (py::ssize_t)4});
+    std::memcpy(seeds.mutable_data(), k.seeds.data(), k.n_keys * 16);
+    d["seeds"] = seeds;
+    for (int s = 0; s < 2; ++s) {
+      py::array_t<std::int32_t> cw({(py::ssize_t)k.n_codewords, (py::ssize_t)4});
+      std::memcpy(cw.mutable_data(), k.cw[s].data(), k.n_codewords * 16);
+      d[s == 0 ? "cw_even" : "cw_odd"] = cw;
+    }
+    d["n_keys"] = (long long)k.n_keys;
+    d["n_codewords"] = (long long)k.n_codewords;
+    return d;
+  };
+  return py::make_tuple(pack(k0), pack(k1));
+}
+
+py::array_t<std::int32_t> grid_expand(py::dict key, int prf_method) {
+  GridDpfKey k;
+  k.n_keys = (u64)py::cast<long long>(key["n_keys"]);
+  k.n_codewords = (u64)py::cast<long long>(key["n_codewords"]);
+  auto seeds = py::cast<py::array_t<std::int32_t>>(key["seeds"]);
+  auto cwe = py::cast<py::array_t<std::int32_t>>(key["cw_even"]);
+  auto cwo = py::cast<py::array_t<std::int32_t>>(key["cw_odd"]);
+  k.seeds.resize(k.n_keys);
+  k.cw[0].resize(k.n_codewords);
+  k.cw[1].resize(k.n_codewords);
+  std::memcpy(k.seeds.data(), seeds.data(), k.n_keys * 16);
+  std::memcpy(k.cw[0].data(), cwe.data(), k.n_codewords * 16);
+  std::memcpy(k.cw[1].data(), cwo.data(), k.n_codewords * 16);
+  const u64 n = k.n_keys * k.n_codewords;
+  py::array_t<std::int32_t> out((py::ssize_t)n);
+  u32* optr = reinterpret_cast<u32*>(out.mutable_data());
+  {
+    py::gil_scoped_release nogil;
+    for (u64 i = 0; i < n; ++i)
+      optr[i] = (u32)grid_dpf_eval(k, i, prf_method);
+  }
+  return out;
+}
+
 // AES GPU tables (5 x 256 u32: te0..te3, sbox) for upload to the device.
 py::array_t<std::int32_t> aes_gpu_tables() {
   py::array_t<std::int32_t> out(5 * 256);
@@ -200,6 +248,9 @@ PYBIND11_MODULE(_core, m) {
   m.def("shard_subkey", &shard_subkey);
   m.def("aes_gpu_tables", &aes_gpu_tables);
   m.def("gemm128_cpu", &gemm128_cpu);
+  m.def("grid_gen", &grid_gen, py::arg("alpha"), py::arg("n_keys"),
+        py::arg("n_codewords"), py::arg("seed"), py::arg("prf_method"));
+  m.def("grid_expand", &grid_expand, py::arg("key"), py::arg("prf_method"));
   m.attr("KEY_INTS") = py::int_(kKeyInts);
   m.attr("ENTRY_WORDS") = py::int_(kEntryWords);
   m.attr("PRF_DUMMY") = py::int_((int)PRF_DUMMY);

@@ -223,6 +223,48 @@ void key_deserialize(const std::int32_t in[kKeyInts], DpfKey& k) {
 }
 
 // ---------------------------------------------------------------------------
+// sqrt(n) grid construction
+// ---------------------------------------------------------------------------
+void grid_dpf_gen(u64 alpha, u128 beta, u64 n_keys, u64 n_codewords,
+                  int prf_method, KeyRng& rng, GridDpfKey& k0, GridDpfKey& k1) {
+  const u64 n = n_keys * n_codewords;
+  if (alpha >= n) throw std::invalid_argument("alpha must be < n");
+  const u64 jt = alpha % n_keys;       // target column
+  const u64 it = alpha / n_keys;       // target row
+  k0 = GridDpfKey{}; k1 = GridDpfKey{};
+  k0.n_keys = k1.n_keys = n_keys;
+  k0.n_codewords = k1.n_codewords = n_codewords;
+  k0.seeds.resize(n_keys); k1.seeds.resize(n_keys);
+  for (int s = 0; s < 2; ++s) { k0.cw[s].resize(n_codewords); k1.cw[s].resize(n_codewords); }
+
+  for (u64 j = 0; j < n_keys; ++j) {
+    if (j == jt) {
+      k0.seeds[j] = rng.next_u128() & ~(u128)1;  // forced even
+      k1.seeds[j] = rng.next_u128() | 1;         // forced odd
+    } else {
+      k0.seeds[j] = k1.seeds[j] = rng.next_u128();
+    }
+  }
+  const u128 s0 = k0.seeds[jt], s1 = k1.seeds[jt];
+  const int sel0 = (int)(s0 & 1);
+  for (u64 i = 0; i < n_codewords; ++i) {
+    u128 p0 = prf_eval(prf_method, s0, (u128)i);
+    u128 p1 = prf_eval(prf_method, s1, (u128)i);
+    u128 rnd = rng.next_u128();
+    u128 target = (i == it) ? beta : (u128)0;
+    u128 other = p0 + rnd - p1 - target;
+    k0.cw[sel0][i] = k1.cw[sel0][i] = rnd;
+    k0.cw[sel0 ^ 1][i] = k1.cw[sel0 ^ 1][i] = other;
+  }
+}
+
+u128 grid_dpf_eval(const GridDpfKey& k, u64 idx, int prf_method) {
+  const u128 seed = k.seeds[idx % k.n_keys];
+  const u64 row = idx / k.n_keys;
+  return prf_eval(prf_method, seed, (u128)row) + k.cw[(size_t)(seed & 1)][row];
+}
+
+// ---------------------------------------------------------------------------
 // Layout contract
 // ---------------------------------------------------------------------------
 int zlog_for_depth(int depth) {

@@ -22,6 +22,7 @@
 #include <cstring>
 #include <random>
 #include <stdexcept>
+#include <vector>
 
 namespace gpudpf {
 
@@ -134,6 +135,24 @@ u64 leaf_perm_inv(u64 n, int zlog, u64 row);   // permuted row -> natural
 // levels from the root consuming the bits of r LSB-first.
 void dpf_shard_subkey(const DpfKey& k, int prf_method, u64 rank, u64 world,
                       DpfKey& out);
+
+// ---------------------------------------------------------------------------
+// sqrt(n) GRID construction (the reference's general n_keys x n_codewords
+// DPF, dpf_base/dpf.h:290-360): one seed per column, one codeword pair
+// per row; eval(idx) = PRF(seed[idx % n_keys], idx / n_keys)
+//                      + cw[seed parity][idx / n_keys].
+// Kept as a standalone research component (the log-n scheme uses it only
+// conceptually as its N=2 base case here).
+// ---------------------------------------------------------------------------
+struct GridDpfKey {
+  u64 n_keys = 0, n_codewords = 0;
+  std::vector<u128> seeds;      // [n_keys]
+  std::vector<u128> cw[2];      // [n_codewords] each; selected by parity
+};
+
+void grid_dpf_gen(u64 alpha, u128 beta, u64 n_keys, u64 n_codewords,
+                  int prf_method, KeyRng& rng, GridDpfKey& k0, GridDpfKey& k1);
+u128 grid_dpf_eval(const GridDpfKey& k, u64 idx, int prf_method);
 
 inline int ilog2_u64(u64 v) {
   int l = 0;

@@ -137,3 +137,19 @@ def test_shard_subkey(world):
         # rank owns natural rows idx % world == rank, local index idx//world
         assert np.array_equal(e1, full1[rank::world])
         assert np.array_equal(e2, full2[rank::world])
+
+
+@pytest.mark.parametrize("prf", [_core.PRF_DUMMY, _core.PRF_AES128])
+def test_grid_sqrt_n_construction(prf):
+    n_keys, n_codewords = 64, 32
+    n = n_keys * n_codewords
+    alpha = 777
+    k0, k1 = _core.grid_gen(alpha, n_keys, n_codewords, b"grid-seed", prf)
+    a = _core.grid_expand(k0, prf).astype(np.int64)
+    b = _core.grid_expand(k1, prf).astype(np.int64)
+    rec = (a - b).astype(np.int32)
+    assert rec[alpha] == 1
+    assert np.count_nonzero(rec) == 1
+    # communication size: n_keys seeds + 2*n_codewords codewords
+    assert k0["seeds"].shape == (n_keys, 4)
+    assert k0["cw_even"].shape == (n_codewords, 4)
