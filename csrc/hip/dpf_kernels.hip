@@ -386,18 +386,6 @@ __device__ __forceinline__ void expand_pair(uint4 seed, int i,
   c1 = add128(p1, cw_lds[sel * 64 + i * 2 + 1]);
 }
 
-template <int PRF>
-__device__ __forceinline__ void expand_leaf_low(uint4 seed,
-                                                const uint4* cw_lds,
-                                                const AesLds& T, u32& v0,
-                                                u32& v1) {
-  u32 p0, p1;
-  prf_pair_low<PRF>(seed, T, p0, p1);
-  const int sel = (int)(seed.x & 1u);
-  v0 = p0 + cw_lds[sel * 64 + 0].x;
-  v1 = p1 + cw_lds[sel * 64 + 1].x;
-}
-
 // ---------------------------------------------------------------------------
 // Main kernel.  LDS map (u32 granularity):
 //   [cw: 128 uint4][shared: Z*max(2, DS-3) uint4][aes: 0/8192 u32][red]
@@ -444,7 +432,7 @@ __global__ __launch_bounds__(256) void dpf_eval_kernel(
       for (int c = 0; c < AES_REP; ++c) aes_lds[e * AES_REP + c] = v;
     }
   }
-  AesLds T{aes_lds, (u32)(t & 31)};
+  AesLds T{aes_lds, (u32)(t & (AES_REP - 1))};
   uint4* pp = shared_region;
   if (t == 0) {
     const int* rp = keys + key_base + 516;
@@ -494,6 +482,11 @@ __global__ __launch_bounds__(256) void dpf_eval_kernel(
 #pragma unroll
   for (int w = 0; w < 16; ++w) acc[w] = 0;
 
+  // leaf-level correction words (eval level 0) are read every iteration:
+  // hoist their low words into registers
+  const u32 cwl[2][2] = {{cw_lds[0].x, cw_lds[1].x},
+                         {cw_lds[64].x, cw_lds[65].x}};
+
   for (long long j = j_lo; j < j_hi; ++j) {
     // Issue the two table-row loads first: the leaf ciphers below hide
     // their latency.
@@ -510,7 +503,13 @@ __global__ __launch_bounds__(256) void dpf_eval_kernel(
     }
 
     u32 v0, v1;
-    expand_leaf_low<PRF>(cur, cw_lds, T, v0, v1);
+    {
+      u32 p0, p1;
+      prf_pair_low<PRF>(cur, T, p0, p1);
+      const int sel = (int)(cur.x & 1u);
+      v0 = p0 + cwl[sel][0];
+      v1 = p1 + cwl[sel][1];
+    }
 
     if constexpr (FUSED) {
 #pragma unroll
