@@ -182,21 +182,37 @@ class DPF(object):
         res = torch.cat(results) if len(results) > 1 else results[0]
         return res.cpu()
 
-    def _eval_gpu_two_stage(self, keys):
+    def _eval_gpu_two_stage(self, keys, chunk=128):
         """Expand one-hot shares (permuted rows), then reduce against the
         permuted table with the MFMA mod-2^32 GEMM — no gather needed
-        because both sides share the leaf_perm row order."""
+        because both sides share the leaf_perm row order.  Expansion of
+        chunk i+1 overlaps the matmul of chunk i on two HIP streams (the
+        reference's dual-stream expansion||GEMM pattern,
+        dpf_benchmark.cu:191-231)."""
         from gpudpf import ops
 
         kt, n, depth = self._keys_tensor(keys)
         dev = self._table_gpu.device
-        stream = torch.cuda.current_stream(dev).cuda_stream
         keys_gpu = kt.to(dev, non_blocking=True).contiguous()
         b = kt.shape[0]
-        shares = torch.empty((b, n), dtype=torch.int32, device=dev)
-        _hip.eval_expand(keys_gpu.data_ptr(), shares.data_ptr(), self._aes_ptr,
-                         b, n, depth, self._zlog, self.prf_method, stream)
-        out = ops.pir_matmul_u32(shares, self._table_gpu)
+        s_expand = torch.cuda.Stream(dev)
+        s_matmul = torch.cuda.Stream(dev)
+        outs = []
+        for lo in range(0, b, chunk):
+            hi = min(b, lo + chunk)
+            with torch.cuda.stream(s_expand):
+                shares = torch.empty((hi - lo, n), dtype=torch.int32, device=dev)
+                _hip.eval_expand(
+                    keys_gpu[lo:hi].contiguous().data_ptr(), shares.data_ptr(),
+                    self._aes_ptr, hi - lo, n, depth, self._zlog,
+                    self.prf_method, s_expand.cuda_stream)
+            s_matmul.wait_stream(s_expand)
+            shares.record_stream(s_matmul)
+            with torch.cuda.stream(s_matmul):
+                outs.append(ops.pir_matmul_u32(shares, self._table_gpu))
+        torch.cuda.current_stream(dev).wait_stream(s_matmul)
+        torch.cuda.current_stream(dev).wait_stream(s_expand)
+        out = torch.cat(outs) if len(outs) > 1 else outs[0]
         return out[:, : self.table_effective_entry_size].cpu()
 
     def eval_gpu_into(self, keys_gpu, out_gpu):
