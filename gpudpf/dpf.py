@@ -51,56 +51,67 @@ class DPF(object):
         self._perm_gpu = None           # natural->row map (for one-hot unperm)
         self._zlog = None
         self._depth = None
+        self._n_domain = None
+        self._entry_padded = None
         self._aes_ptr = 0
 
     # ------------------------------------------------------------------
     # Client side
     # ------------------------------------------------------------------
+    @staticmethod
+    def _domain(n):
+        """DPF tree domain for an n-entry table: next power of two >= 128
+        (non-power-of-two tables are zero-padded — a capability the
+        reference leaves as a TODO, dpf.py:24)."""
+        d = 128
+        while d < n:
+            d <<= 1
+        return d
+
     def gen(self, k, n):
         """Generate the two server keys selecting index k of an n-entry
-        table.  Returns two int32[524] torch tensors (2096-byte keys)."""
-        if n & (n - 1) != 0:
-            raise Exception("Table num entries (%d) must be a power of two" % n)
+        table (any n >= 1; non-powers of two use the next power-of-two
+        domain).  Returns two int32[524] torch tensors (2096-byte keys)."""
         if k >= n:
             raise Exception(
                 "k (%d), the selected element, must be less than n (%d), the "
                 "number of entries in the table" % (k, n)
             )
         seed = os.urandom(128)
-        k1, k2 = _core.gen(k, n, seed, self.prf_method)
+        k1, k2 = _core.gen(k, self._domain(n), seed, self.prf_method)
         return [torch.from_numpy(k1), torch.from_numpy(k2)]
 
     # ------------------------------------------------------------------
     # Server side
     # ------------------------------------------------------------------
     def eval_init(self, table):
-        """Upload an [n, e] int32 table (n power-of-two >= 128, e <= 16).
-        Rows are reordered by the kernel layout contract (leaf_perm) and
-        padded to 16 words."""
+        """Upload an [n, e] int32 table.  n is padded to the next
+        power-of-two domain (>= 128) and e to a multiple of 16; rows are
+        reordered by the kernel layout contract (leaf_perm).  Entries
+        wider than 16 words are served by the two-stage MFMA path."""
         self.table = table
         self.table_num_entries = int(table.shape[0])
         self.table_effective_entry_size = int(table.shape[1])
         n, e = self.table_num_entries, self.table_effective_entry_size
 
-        if n < 128:
-            raise Exception("Table (%d) must have at least 128 elements" % n)
-        if n & (n - 1) != 0:
-            raise Exception("Table num entries (%d) must be a power of two" % n)
-        if e > self.ENTRY_SIZE:
-            raise Exception(
-                "Table entry dimension (%d) must be < %d" % (e, self.ENTRY_SIZE)
-            )
+        # extensions over the reference: non-power-of-two n (zero-padded to
+        # the next power-of-two domain) and entries wider than 16 words
+        # (served by the two-stage MFMA path; reference TODOs dpf.py:16-24)
+        nd = self._domain(n)
+        ep = -(-e // self.ENTRY_SIZE) * self.ENTRY_SIZE
+        self._n_domain = nd
+        self._entry_padded = ep
 
-        self._depth = n.bit_length() - 1
+        self._depth = nd.bit_length() - 1
         self._zlog = _core.zlog_for_depth(self._depth)
 
         if self.device is None:
             self.device = "cuda:0" if torch.cuda.is_available() else "cpu"
         dev = torch.device(self.device)
 
-        padded = torch.zeros((n, self.ENTRY_SIZE), dtype=torch.int32)
-        padded[:, :e] = table.to(torch.int32)
-        perm = torch.from_numpy(_core.leaf_perm_table(n, self._zlog))
+        padded = torch.zeros((nd, ep), dtype=torch.int32)
+        padded[:n, :e] = table.to(torch.int32)
+        perm = torch.from_numpy(_core.leaf_perm_table(nd, self._zlog))
         reordered = torch.empty_like(padded)
         reordered[perm] = padded  # row perm[i] <- natural row i
 
@@ -139,18 +150,22 @@ class DPF(object):
         strategy: "fused" (production: expansion fused with the table MAC)
         or "two_stage" (expand one-hot shares, then multiply against the
         table on the MFMA matrix cores — the runtime strategy selection the
-        reference leaves as a TODO, dpf.py:26)."""
-        if strategy == "two_stage" and not one_hot_only:
-            return self._eval_gpu_two_stage(keys)
+        reference leaves as a TODO, dpf.py:26).  Entries wider than 16
+        words route through "two_stage" automatically."""
         if self._table_gpu is None:
             raise Exception("Must call `eval_init` before `eval_gpu`")
         if not _HAS_HIP:
             raise Exception("gpudpf._hip extension is not available")
         kt, n, depth = self._keys_tensor(keys)
-        if n != self.table_num_entries:
+        if n != self._n_domain:
             raise Exception(
-                "key domain (%d) does not match table (%d)" % (n, self.table_num_entries)
+                "key domain (%d) does not match table domain (%d)"
+                % (n, self._n_domain)
             )
+        if self._entry_padded > self.ENTRY_SIZE and not one_hot_only:
+            strategy = "two_stage"  # wide entries go through the MFMA path
+        if strategy == "two_stage" and not one_hot_only:
+            return self._eval_gpu_two_stage(kt)
         batch = kt.shape[0]
         dev = self._table_gpu.device
         stream = torch.cuda.current_stream(dev).cuda_stream
@@ -167,8 +182,10 @@ class DPF(object):
                     chunk.data_ptr(), out.data_ptr(), self._aes_ptr, b, n,
                     depth, self._zlog, self.prf_method, stream,
                 )
-                # rows are in leaf_perm order; gather back to natural order
+                # rows are in leaf_perm order; gather back to natural
+                # order and trim the power-of-two padding
                 out = out.index_select(1, self._perm_gpu)
+                out = out[:, : self.table_num_entries]
             else:
                 # zeroed: the kernel's j-split segments accumulate with atomics
                 out = torch.zeros((b, self.ENTRY_SIZE), dtype=torch.int32, device=dev)
@@ -245,9 +262,11 @@ class DPF(object):
             raise Exception(
                 "Must call `eval_init` before `eval_cpu` with one_hot_only=False"
             )
-        return torch.matmul(one_hots.to(torch.int64), self.table.to(torch.int64)).to(
-            torch.int32
-        )
+        # domain padding rows carry zero table entries: slice them off
+        n = self.table.shape[0]
+        return torch.matmul(
+            one_hots[:, :n].to(torch.int64), self.table.to(torch.int64)
+        ).to(torch.int32)
 
     def __repr__(self):
         if self.table is None:

@@ -45,19 +45,23 @@ def test_cpu_dpf_with_table():
 def test_gen_validation():
     dpf = DPF()
     with pytest.raises(Exception):
-        dpf.gen(0, 1000)  # not a power of two
-    with pytest.raises(Exception):
         dpf.gen(1024, 1024)  # k >= n
+    # non-power-of-two n is supported (padded domain); key domain is the
+    # next power of two >= 128
+    k1, _ = dpf.gen(0, 1000)
+    assert k1[520] == 1024
+    k1, _ = dpf.gen(0, 50)
+    assert k1[520] == 128
 
 
-def test_eval_init_validation():
+def test_eval_init_accepts_extended_shapes():
     dpf = DPF(device="cpu")
-    with pytest.raises(Exception):
-        dpf.eval_init(torch.zeros((64, 16), dtype=torch.int32))  # < 128 rows
-    with pytest.raises(Exception):
-        dpf.eval_init(torch.zeros((100, 16), dtype=torch.int32))  # not pow2
-    with pytest.raises(Exception):
-        dpf.eval_init(torch.zeros((128, 17), dtype=torch.int32))  # entry too big
+    dpf.eval_init(torch.zeros((64, 16), dtype=torch.int32))   # < 128 rows: padded
+    assert dpf._n_domain == 128
+    dpf.eval_init(torch.zeros((100, 16), dtype=torch.int32))  # not pow2: padded
+    assert dpf._n_domain == 128
+    dpf.eval_init(torch.zeros((128, 17), dtype=torch.int32))  # wide entry
+    assert dpf._entry_padded == 32
 
 
 def test_key_size_constant():
@@ -81,3 +85,29 @@ def test_repr():
     assert "uninitialized" in repr(dpf)
     dpf.eval_init(torch.zeros((128, 4), dtype=torch.int32))
     assert "entries=128" in repr(dpf) and "entry_size=4" in repr(dpf)
+
+
+def test_non_pow2_table_cpu():
+    N = 1000  # padded to 1024 domain
+    dpf = DPF(prf=DPF.PRF_SALSA20, device="cpu")
+    table = torch.arange(N * 4, dtype=torch.int32).reshape(N, 4)
+    dpf.eval_init(table)
+    idxs = [0, 999, 500]
+    k1s, k2s = [], []
+    for i in idxs:
+        k1, k2 = dpf.gen(i, N)
+        k1s.append(k1)
+        k2s.append(k2)
+    rec = (dpf.eval_cpu(k1s).to(torch.int64) -
+           dpf.eval_cpu(k2s).to(torch.int64)).to(torch.int32)
+    assert torch.equal(rec, table[idxs, :])
+
+
+def test_small_table_padded_to_128():
+    dpf = DPF(device="cpu")
+    table = torch.ones((50, 2), dtype=torch.int32)
+    dpf.eval_init(table)
+    k1, k2 = dpf.gen(10, 50)
+    rec = (dpf.eval_cpu([k1]).to(torch.int64) -
+           dpf.eval_cpu([k2]).to(torch.int64)).to(torch.int32)
+    assert torch.equal(rec[0], table[10])

@@ -179,3 +179,25 @@ def test_gpu_two_stage_strategy_matches_fused():
         assert torch.equal(fused, two), prf
         rec = (two[0].to(torch.int64) - two[1].to(torch.int64)).to(torch.int32)
         assert torch.equal(rec, table[alpha])
+
+
+def test_gpu_wide_entries_and_non_pow2():
+    # entries wider than 16 words route through the MFMA two-stage path;
+    # non-power-of-two n pads to the next domain
+    n, e = 5000, 40
+    dpf = DPF(prf=DPF.PRF_CHACHA20)
+    table = torch.randint(-(2**31), 2**31 - 1, (n, e), dtype=torch.int64).to(
+        torch.int32
+    )
+    dpf.eval_init(table)
+    idxs = [0, 4999, 1234]
+    k1s, k2s = [], []
+    for i in idxs:
+        k1, k2 = dpf.gen(i, n)
+        k1s.append(k1)
+        k2s.append(k2)
+    a = dpf.eval_gpu(k1s)
+    b = dpf.eval_gpu(k2s)
+    assert a.shape == (3, e)
+    rec = (a.to(torch.int64) - b.to(torch.int64)).to(torch.int32)
+    assert torch.equal(rec, table[idxs, :])
