@@ -49,6 +49,9 @@ class ShardedDPF(object):
         """Initialize from the FULL table (each rank slices its residue
         class).  For tables too large to materialize, build the shard
         directly and call eval_init_local."""
+        n = int(table.shape[0])
+        if n & (n - 1) != 0:
+            raise Exception("sharded tables must have power-of-two entries")
         self.table_num_entries = int(table.shape[0])
         self.table_effective_entry_size = int(table.shape[1])
         self.local.eval_init(self.shard_rows(table).contiguous())

@@ -201,3 +201,19 @@ def test_gpu_wide_entries_and_non_pow2():
     assert a.shape == (3, e)
     rec = (a.to(torch.int64) - b.to(torch.int64)).to(torch.int32)
     assert torch.equal(rec, table[idxs, :])
+
+
+def test_gpu_fused_deterministic():
+    # the j-split combines segment partials with atomics; mod-2^32 addition
+    # is commutative/associative, so results must be bitwise reproducible
+    n = 1 << 17
+    dpf = DPF(prf=DPF.PRF_SALSA20)
+    k1, _ = dpf.gen(4242, n)
+    keys = torch.stack([k1] * 64)
+    table = torch.randint(-(2**31), 2**31 - 1, (n, 16), dtype=torch.int64).to(
+        torch.int32
+    )
+    dpf.eval_init(table)
+    a = dpf.eval_gpu(keys)
+    for _ in range(3):
+        assert torch.equal(dpf.eval_gpu(keys), a)
