@@ -47,8 +47,9 @@ def test_gpu_fused_sweep():
                    seed=n)
 
 
-def test_gpu_fused_large_batch_chunking():
-    # batch larger than BATCH_SIZE exercises chunking
+def test_gpu_fused_large_batch_chunking(monkeypatch):
+    # force multiple kernel launches per eval_gpu call
+    monkeypatch.setattr(DPF, "MAX_LAUNCH_BATCH", 256)
     _roundtrip(1024, 600, 5, DPF.PRF_CHACHA20)
 
 
@@ -217,3 +218,27 @@ def test_gpu_fused_deterministic():
     a = dpf.eval_gpu(keys)
     for _ in range(3):
         assert torch.equal(dpf.eval_gpu(keys), a)
+
+
+def test_gpu_graphed_server_matches_eval():
+    from gpudpf.serving import GraphedServer
+
+    n, batch = 16384, 64
+    dpf = DPF(prf=DPF.PRF_AES128)
+    table = torch.randint(-(2**31), 2**31 - 1, (n, 16), dtype=torch.int64).to(
+        torch.int32
+    )
+    dpf.eval_init(table)
+    srv = GraphedServer(dpf, batch)
+    keys = []
+    for i in range(batch):
+        k1, _ = dpf.gen((i * 37) % n, n)
+        keys.append(k1)
+    kt = torch.stack(keys)
+    want = dpf.eval_gpu(kt)
+    got = srv.eval(kt)
+    assert torch.equal(got, want)
+    # replay with different keys reuses the same graph
+    keys2 = [dpf.gen((i * 11) % n, n)[0] for i in range(batch)]
+    kt2 = torch.stack(keys2)
+    assert torch.equal(srv.eval(kt2), dpf.eval_gpu(kt2))
