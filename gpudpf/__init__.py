@@ -15,6 +15,6 @@ Public API parity (reference dpf.py:35-137):
 """
 
 from gpudpf.dpf import DPF  # noqa: F401
-from gpudpf.dist import ShardedDPF  # noqa: F401
+from gpudpf.dist import ReplicatedDPF, ShardedDPF  # noqa: F401
 
 __version__ = "0.1.0"
