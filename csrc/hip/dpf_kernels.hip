@@ -41,6 +41,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <mutex>
 #include <stdexcept>
 #include <string>
 
@@ -85,6 +86,9 @@ __device__ __forceinline__ u32 rotr24(u32 v) { return (v >> 24) | (v << 8); }
 
 // AES replicated-table geometry
 #define AES_REP 32
+// DFS sibling-stack levels kept in LDS (deeper levels: registers;
+// shallower: global scratch)
+#define MAX_LDS_LEVELS 4
 #define AES_LDS_WORDS (256 * AES_REP)  // 32 KiB
 
 // ---------------------------------------------------------------------------
@@ -405,20 +409,30 @@ __device__ __forceinline__ void expand_pair(uint4 seed, int i,
 template <int PRF, bool FUSED>
 __global__ __launch_bounds__(256) void dpf_eval_kernel(
     const int* __restrict__ keys, const u32* __restrict__ table,
-    u32* __restrict__ out, const u32* __restrict__ aes_tabs, int depth,
-    int zlog, int slog, long long n) {
+    u32* __restrict__ out, const u32* __restrict__ aes_tabs,
+    uint4* __restrict__ scratch, int depth, int zlog, int slog,
+    long long n) {
   extern __shared__ u32 smem[];
   const int Z = 1 << zlog;
   const int DS = depth - zlog;  // subtree splits per thread (>= 1)
-  const int lds_levels = DS > 3 ? DS - 3 : 0;
+  // Sibling-stack placement by pop frequency (level d pops once per
+  // 2^(DS-1-d) pairs): deepest two levels in registers, the next
+  // MAX_LDS_LEVELS in LDS, the shallow remainder in global scratch
+  // (touched <= 1/64 of steps) — keeping LDS small enough for 3+
+  // workgroups per CU even with the 32 KB AES table resident.
+  const int lds_levels =
+      DS > 3 ? (DS - 3 < MAX_LDS_LEVELS ? DS - 3 : MAX_LDS_LEVELS) : 0;
+  const int lds_base = DS - 2 - lds_levels;  // levels [lds_base, DS-3] in LDS
+  const int glob_levels = (DS - 3) - lds_levels;  // levels [1, lds_base-1]
   const int t = (int)threadIdx.x;
   const int key_id = (int)(blockIdx.x >> slog);
   const int seg = (int)(blockIdx.x & ((1u << slog) - 1));
   const long long key_base = (long long)key_id * 524;
 
   uint4* cw_lds = reinterpret_cast<uint4*>(smem);   // 128 entries
-  uint4* shared_region = cw_lds + 128;              // Z*max(2, DS-3)
+  uint4* shared_region = cw_lds + 128;
   const int shared_u4 = Z * (lds_levels > 2 ? lds_levels : 2);
+  uint4* gstack = scratch + (size_t)blockIdx.x * (glob_levels > 0 ? glob_levels : 0) * Z;
   u32* aes_lds = reinterpret_cast<u32*>(shared_region + shared_u4);
   u32* red = aes_lds + (PRF == PRF_AES128 ? AES_LDS_WORDS : 0);
 
@@ -459,7 +473,7 @@ __global__ __launch_bounds__(256) void dpf_eval_kernel(
   }
   uint4 cur = a[t];
   __syncthreads();  // everyone holds their frontier seed; pp is now free
-  uint4* stack = shared_region;  // cold stack levels 1..DS-3 (slot d-1)
+  uint4* stack = shared_region;  // LDS stack levels [lds_base, DS-3]
 
   // Targeted descent to leaf-pair j_lo: at split d take bit (DS-1-d) of
   // j_lo, always recording the bit-1 child in the level's sibling slot
@@ -474,7 +488,8 @@ __global__ __launch_bounds__(256) void dpf_eval_kernel(
     expand_pair<PRF>(cur, DS - d, cw_lds, T, c0, c1);
     if (d == DS - 1) rtop1 = c1;
     else if (d == DS - 2) rtop2 = c1;
-    else stack[(d - 1) * Z + t] = c1;
+    else if (d >= lds_base) stack[(d - lds_base) * Z + t] = c1;
+    else gstack[(size_t)(d - 1) * Z + t] = c1;
     cur = ((j_lo >> (DS - 1 - d)) & 1) ? c1 : c0;
   }
 
@@ -535,13 +550,16 @@ __global__ __launch_bounds__(256) void dpf_eval_kernel(
       continue;  // next leaf-parent is the sibling itself
     }
     const int dpop = DS - 1 - c;
-    cur = (c == 1) ? rtop2 : stack[(dpop - 1) * Z + t];
+    cur = (c == 1) ? rtop2
+          : (dpop >= lds_base) ? stack[(dpop - lds_base) * Z + t]
+                               : gstack[(size_t)(dpop - 1) * Z + t];
     for (int d = dpop + 1; d <= DS - 1; ++d) {
       uint4 c0, c1;
       expand_pair<PRF>(cur, DS - d, cw_lds, T, c0, c1);
       if (d == DS - 1) rtop1 = c1;
       else if (d == DS - 2) rtop2 = c1;
-      else stack[(d - 1) * Z + t] = c1;
+      else if (d >= lds_base) stack[(d - lds_base) * Z + t] = c1;
+      else gstack[(size_t)(d - 1) * Z + t] = c1;
       cur = c0;
     }
   }
@@ -677,12 +695,39 @@ __global__ __launch_bounds__(256) void dpf_naive_kernel(
 namespace {
 
 size_t fused_shmem_bytes(int Z, int DS, int prf) {
-  const int lds_levels = DS > 3 ? DS - 3 : 0;
+  int lds_levels = DS > 3 ? DS - 3 : 0;
+  if (lds_levels > MAX_LDS_LEVELS) lds_levels = MAX_LDS_LEVELS;
   size_t bytes = 128 * 16;                                  // codewords
   bytes += (size_t)Z * (lds_levels > 2 ? lds_levels : 2) * 16;  // pp/stack
   if (prf == PRF_AES128) bytes += AES_LDS_WORDS * 4;        // replicated te0
   bytes += (size_t)(Z / 64) * 16 * 4;                       // reduction
   return bytes;
+}
+
+// Grow-only per-device scratch for the global stack levels.  Reused
+// across launches; sized for the largest request seen.  NOTE: concurrent
+// fused/expand launches on DIFFERENT streams of one device would share
+// this buffer — the python API serializes launches per DPF call, which
+// is the supported pattern.
+std::mutex g_scratch_mu;
+void* g_scratch[64] = {};
+size_t g_scratch_bytes[64] = {};
+
+uint4* get_scratch(size_t bytes, hipStream_t stream) {
+  if (bytes == 0) return nullptr;
+  int dev = 0;
+  HIP_CHECK(hipGetDevice(&dev));
+  std::lock_guard<std::mutex> lock(g_scratch_mu);
+  if (g_scratch_bytes[dev] < bytes) {
+    if (g_scratch[dev]) {
+      HIP_CHECK(hipStreamSynchronize(stream));
+      HIP_CHECK(hipFree(g_scratch[dev]));
+      g_scratch[dev] = nullptr;
+    }
+    HIP_CHECK(hipMalloc(&g_scratch[dev], bytes));
+    g_scratch_bytes[dev] = bytes;
+  }
+  return reinterpret_cast<uint4*>(g_scratch[dev]);
 }
 
 template <int PRF, bool FUSED>
@@ -705,9 +750,14 @@ void launch_eval_t(const int* keys, const u32* table, u32* out,
                                   hipFuncAttributeMaxDynamicSharedMemorySize,
                                   (int)shmem));
   }
+  int lds_levels = DS > 3 ? DS - 3 : 0;
+  if (lds_levels > MAX_LDS_LEVELS) lds_levels = MAX_LDS_LEVELS;
+  const int glob_levels = (DS > 3 ? DS - 3 : 0) - lds_levels;
+  uint4* scratch = get_scratch(
+      (size_t)(batch << slog) * glob_levels * Z * 16, stream);
   hipLaunchKernelGGL(kern, dim3((unsigned)(batch << slog)), dim3((unsigned)Z),
-                     shmem, stream, keys, table, out, aes_tabs, depth, zlog,
-                     slog, n);
+                     shmem, stream, keys, table, out, aes_tabs, scratch,
+                     depth, zlog, slog, n);
   HIP_CHECK(hipGetLastError());
 }
 
