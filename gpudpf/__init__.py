@@ -10,7 +10,7 @@ Public API parity (reference dpf.py:35-137):
     from gpudpf import DPF
     d = DPF(prf=DPF.PRF_AES128)
     k1, k2 = d.gen(k, n)
-    d.eval_init(table)          # [n, e<=16] int32, n a power of two >= 128
+    d.eval_init(table)          # [n, e] int32 (any n; e<=16 uses the fused path)
     shares = d.eval_gpu([k1])   # [1, e] int32 secret shares
 """
 

@@ -47,7 +47,7 @@ class DPF(object):
         self.table = None               # natural-order table (as given)
         self.table_num_entries = None
         self.table_effective_entry_size = None
-        self._table_gpu = None          # leaf_perm-reordered [n,16] int32 on GPU
+        self._table_gpu = None          # leaf_perm-reordered [nd,ep] int32 on GPU
         self._perm_gpu = None           # natural->row map (for one-hot unperm)
         self._zlog = None
         self._depth = None
@@ -242,7 +242,7 @@ class DPF(object):
         out_gpu.zero_()  # j-split segments accumulate with atomics
         _hip.eval_fused(
             keys_gpu.data_ptr(), self._table_gpu.data_ptr(), out_gpu.data_ptr(),
-            self._aes_ptr, b, self.table_num_entries, self._depth, self._zlog,
+            self._aes_ptr, b, self._n_domain, self._depth, self._zlog,
             self.prf_method, stream,
         )
 

@@ -35,7 +35,7 @@ import collections
 import hashlib
 import json
 import os
-from typing import Dict, List, NamedTuple, Optional, Sequence
+from typing import Dict, NamedTuple, Optional, Sequence
 
 
 class HotColdConfig(NamedTuple):
