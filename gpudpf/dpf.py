@@ -109,15 +109,17 @@ class DPF(object):
             self.device = "cuda:0" if torch.cuda.is_available() else "cpu"
         dev = torch.device(self.device)
 
-        padded = torch.zeros((nd, ep), dtype=torch.int32)
-        padded[:n, :e] = table.to(torch.int32)
         perm = torch.from_numpy(_core.leaf_perm_table(nd, self._zlog))
-        reordered = torch.empty_like(padded)
-        reordered[perm] = padded  # row perm[i] <- natural row i
-
         if dev.type == "cuda":
-            self._table_gpu = reordered.to(dev).contiguous()
+            # pad + permute on the GPU: the row scatter of a multi-GB
+            # table is memory-bandwidth work, not host work
+            padded = torch.zeros((nd, ep), dtype=torch.int32, device=dev)
+            padded[:n, :e] = table.to(torch.int32).to(dev, non_blocking=True)
             self._perm_gpu = perm.to(dev)
+            reordered = torch.empty_like(padded)
+            reordered[self._perm_gpu] = padded  # row perm[i] <- natural row i
+            self._table_gpu = reordered.contiguous()
+            del padded
             if self.prf_method == self.PRF_AES128:
                 self._aes_ptr = _hip.ensure_aes_tables(dev.index or 0)
         else:
