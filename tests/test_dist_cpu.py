@@ -45,11 +45,11 @@ def _worker(rank, world, port, q):
         td.destroy_process_group()
 
 
-@pytest.mark.parametrize("world", [2])
+@pytest.mark.parametrize("world", [2, 4])
 def test_sharded_eval_matches_single(world):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29781
+    port = 29781 + world
     procs = [
         ctx.Process(target=_worker, args=(r, world, port, q)) for r in range(world)
     ]
