@@ -383,11 +383,17 @@ __device__ __forceinline__ void expand_pair(uint4 seed, int i,
                                             const uint4* cw_lds,
                                             const AesLds& T, uint4& c0,
                                             uint4& c1) {
+  // Issue the correction-word reads BEFORE the PRF core: their address
+  // depends only on the seed parity, and the ~600-instruction cipher
+  // hides the LDS latency (left to itself the scheduler sinks them to
+  // +5 instructions before their wait — measured in the .s).
+  const int sel = (int)(seed.x & 1u);
+  uint4 cw0 = cw_lds[sel * 64 + i * 2 + 0];
+  uint4 cw1 = cw_lds[sel * 64 + i * 2 + 1];
   uint4 p0, p1;
   prf_pair<PRF>(seed, T, p0, p1);
-  const int sel = (int)(seed.x & 1u);
-  c0 = add128(p0, cw_lds[sel * 64 + i * 2 + 0]);
-  c1 = add128(p1, cw_lds[sel * 64 + i * 2 + 1]);
+  c0 = add128(p0, cw0);
+  c1 = add128(p1, cw1);
 }
 
 // ---------------------------------------------------------------------------
