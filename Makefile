@@ -33,3 +33,12 @@ profile-pmc: build
 
 clean:
 	rm -f gpudpf/_core*.so gpudpf/_hip*.so
+
+# host-side address sanitizer build of the CPU core (race/UB tier)
+asan:
+	g++ -O1 -g -fsanitize=address -std=c++17 -shared -fPIC -pthread \
+	  -I$$(python -c "import pybind11;print(pybind11.get_include())") \
+	  -I$$(python -c "import sysconfig;print(sysconfig.get_paths()['include'])") \
+	  csrc/core/dpf_core.cc csrc/core/prf.cc csrc/core/aes128.cc \
+	  csrc/core/core_bindings.cc -o /tmp/_core_asan.so
+	@echo "ASan build OK (load with LD_PRELOAD=libasan.so python ...)"
