@@ -41,6 +41,30 @@ py::tuple gen(u64 alpha, u64 n, py::bytes seed, int prf_method) {
   return py::make_tuple(key_to_array(k0), key_to_array(k1));
 }
 
+// Batched key generation: alphas int64[B] -> (k0s, k1s) int32[B,524].
+// One seeded RNG stream covers the whole batch (deterministic).
+py::tuple gen_batch(py::array_t<std::int64_t, py::array::c_style | py::array::forcecast> alphas,
+                    u64 n, py::bytes seed, int prf_method) {
+  std::string sd = seed;
+  KeyRng rng(reinterpret_cast<const unsigned char*>(sd.data()), sd.size());
+  const py::ssize_t b = alphas.size();
+  py::array_t<std::int32_t> k0s({b, (py::ssize_t)kKeyInts});
+  py::array_t<std::int32_t> k1s({b, (py::ssize_t)kKeyInts});
+  const std::int64_t* ap = alphas.data();
+  std::int32_t* p0 = k0s.mutable_data();
+  std::int32_t* p1 = k1s.mutable_data();
+  {
+    py::gil_scoped_release nogil;
+    DpfKey k0, k1;
+    for (py::ssize_t i = 0; i < b; ++i) {
+      dpf_gen((u64)ap[i], 1, n, prf_method, rng, k0, k1);
+      key_serialize(k0, p0 + i * kKeyInts);
+      key_serialize(k1, p1 + i * kKeyInts);
+    }
+  }
+  return py::make_tuple(k0s, k1s);
+}
+
 // Full-domain expansion to low-32 shares, natural order -> int32[n].
 py::array_t<std::int32_t> expand(const KeyArr& key, int prf_method) {
   DpfKey k = key_from_array(key);
@@ -235,6 +259,8 @@ PYBIND11_MODULE(_core, m) {
   m.doc() = "gpudpf CPU core (keygen, reference eval, layout)";
   m.def("gen", &gen, py::arg("alpha"), py::arg("n"), py::arg("seed"),
         py::arg("prf_method"));
+  m.def("gen_batch", &gen_batch, py::arg("alphas"), py::arg("n"),
+        py::arg("seed"), py::arg("prf_method"));
   m.def("expand", &expand, py::arg("key"), py::arg("prf_method"));
   m.def("expand_batch", &expand_batch, py::arg("keys"), py::arg("prf_method"),
         py::arg("num_threads") = 1);

@@ -81,6 +81,18 @@ class DPF(object):
         k1, k2 = _core.gen(k, self._domain(n), seed, self.prf_method)
         return [torch.from_numpy(k1), torch.from_numpy(k2)]
 
+    def gen_batch(self, indices, n):
+        """Generate keys for a batch of secret indices in one call.
+        Returns (k1s, k2s) int32[B,524] tensors."""
+        for k in indices:
+            if k >= n:
+                raise Exception("index %d out of range for n=%d" % (k, n))
+        seed = os.urandom(128)
+        k1s, k2s = _core.gen_batch(
+            np.asarray(indices, dtype=np.int64), self._domain(n), seed,
+            self.prf_method)
+        return torch.from_numpy(k1s), torch.from_numpy(k2s)
+
     # ------------------------------------------------------------------
     # Server side
     # ------------------------------------------------------------------

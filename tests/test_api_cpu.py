@@ -111,3 +111,26 @@ def test_small_table_padded_to_128():
     rec = (dpf.eval_cpu([k1]).to(torch.int64) -
            dpf.eval_cpu([k2]).to(torch.int64)).to(torch.int32)
     assert torch.equal(rec[0], table[10])
+
+
+def test_gen_batch():
+    N = 2048
+    dpf = DPF(prf=DPF.PRF_SALSA20, device="cpu")
+    idxs = [0, 5, 2047, 1024]
+    k1s, k2s = dpf.gen_batch(idxs, N)
+    assert k1s.shape == (4, 524)
+    v1 = dpf.eval_cpu(k1s, one_hot_only=True)
+    v2 = dpf.eval_cpu(k2s, one_hot_only=True)
+    rec = (v1 - v2).numpy()
+    for i, a in enumerate(idxs):
+        assert rec[i, a] == 1 and np.count_nonzero(rec[i]) == 1
+
+
+def test_corrupt_key_rejected():
+    dpf = DPF(device="cpu")
+    garbage = torch.zeros(524, dtype=torch.int32)  # depth 0: invalid
+    with pytest.raises(Exception):
+        dpf.eval_cpu([garbage], one_hot_only=True)
+    garbage[0] = 99  # depth out of range
+    with pytest.raises(Exception):
+        dpf.eval_cpu([garbage], one_hot_only=True)
