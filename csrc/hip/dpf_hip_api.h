@@ -25,6 +25,11 @@ void launch_naive(std::uintptr_t keys, std::uintptr_t out,
                   std::uintptr_t aes_tabs, int batch, long long n, int depth,
                   int prf, std::uintptr_t stream);
 
+// PRF speed-of-light microbenchmark: blocks x 256 threads each run a
+// dependent chain of `iters` pair expansions; out: u32[blocks*256].
+void launch_prf_sol(std::uintptr_t aes_tabs, std::uintptr_t out, int blocks,
+                    int iters, int prf, std::uintptr_t stream);
+
 // Exact u128 GEMM (research harness; gemm128.hip).  a: [M,K] u128 (as
 // 4xint32 limbs), bt: [N,K] u128, c: [M,N] u128, partials: device scratch
 // of gemm128_ksplit(M,N,K)*M*N u128.

@@ -696,6 +696,60 @@ __global__ __launch_bounds__(256) void dpf_naive_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// PRF speed-of-light microbenchmark: dependent chain of pair expansions on
+// register data only (no DFS, no table, no LDS traffic beyond the AES
+// table).  Upper-bounds what any expansion kernel could reach; the fused
+// kernel's pair rate divided by this is its efficiency.
+// ---------------------------------------------------------------------------
+template <int PRF>
+__global__ __launch_bounds__(256) void prf_sol_kernel(
+    const u32* __restrict__ aes_tabs, u32* __restrict__ out, int iters) {
+  extern __shared__ u32 smem[];
+  u32* aes_lds = smem;
+  if constexpr (PRF == PRF_AES128) {
+    for (int e = (int)threadIdx.x; e < 256; e += blockDim.x) {
+      const u32 v = aes_tabs[e];
+#pragma unroll
+      for (int c = 0; c < AES_REP; ++c) aes_lds[e * AES_REP + c] = v;
+    }
+    __syncthreads();
+  }
+  AesLds T{aes_lds, (u32)(threadIdx.x & 31)};
+  uint4 seed = make_uint4(threadIdx.x + 1, blockIdx.x + 2, 3, 4);
+  for (int i = 0; i < iters; ++i) {
+    uint4 r0, r1;
+    prf_pair<PRF>(seed, T, r0, r1);
+    seed = make_uint4(r0.x ^ r1.x, r0.y ^ r1.y, r0.z ^ r1.z, r0.w ^ r1.w);
+  }
+  out[(u64)blockIdx.x * blockDim.x + threadIdx.x] = seed.x;
+}
+
+void launch_prf_sol(std::uintptr_t aes_tabs, std::uintptr_t out, int blocks,
+                    int iters, int prf, std::uintptr_t stream) {
+  auto* a = reinterpret_cast<const u32*>(aes_tabs);
+  auto* o = reinterpret_cast<u32*>(out);
+  auto st = reinterpret_cast<hipStream_t>(stream);
+  const size_t shmem = (prf == PRF_AES128) ? AES_LDS_WORDS * 4 : 0;
+  switch (prf) {
+    case PRF_DUMMY:
+      hipLaunchKernelGGL(prf_sol_kernel<PRF_DUMMY>, dim3(blocks), dim3(256), shmem, st, a, o, iters);
+      break;
+    case PRF_SALSA20:
+      hipLaunchKernelGGL(prf_sol_kernel<PRF_SALSA20>, dim3(blocks), dim3(256), shmem, st, a, o, iters);
+      break;
+    case PRF_CHACHA20:
+      hipLaunchKernelGGL(prf_sol_kernel<PRF_CHACHA20>, dim3(blocks), dim3(256), shmem, st, a, o, iters);
+      break;
+    case PRF_AES128:
+      hipLaunchKernelGGL(prf_sol_kernel<PRF_AES128>, dim3(blocks), dim3(256), shmem, st, a, o, iters);
+      break;
+    default:
+      throw std::invalid_argument("unknown PRF");
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
 // Host launchers
 // ---------------------------------------------------------------------------
 namespace {

@@ -76,6 +76,9 @@ PYBIND11_MODULE(_hip, m) {
         py::arg("k"), py::arg("stream"),
         py::call_guard<py::gil_scoped_release>());
   m.def("gemm128_ksplit", &gpudpf_hip::gemm128_ksplit);
+  m.def("prf_sol", &gpudpf_hip::launch_prf_sol, py::arg("aes_tabs"),
+        py::arg("out"), py::arg("blocks"), py::arg("iters"), py::arg("prf"),
+        py::arg("stream"), py::call_guard<py::gil_scoped_release>());
   m.def("digits", &gpudpf_hip::launch_digits, py::arg("inp"), py::arg("out"),
         py::arg("count"), py::arg("stream"),
         py::call_guard<py::gil_scoped_release>());
