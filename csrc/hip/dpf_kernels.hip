@@ -580,8 +580,6 @@ __global__ __launch_bounds__(256) void dpf_eval_kernel(
       acc[w] = v;
     }
     const int lane = t & 63, wave = t >> 6, nwaves = (int)blockDim.x >> 6;
-    __syncthreads();  // stack region re-used as reduction scratch is NOT —
-                      // red is separate; this sync orders DFS completion
     if (lane == 0) {
 #pragma unroll
       for (int w = 0; w < 16; ++w) red[wave * 16 + w] = acc[w];
