@@ -242,3 +242,24 @@ def test_gpu_graphed_server_matches_eval():
     keys2 = [dpf.gen((i * 11) % n, n)[0] for i in range(batch)]
     kt2 = torch.stack(keys2)
     assert torch.equal(srv.eval(kt2), dpf.eval_gpu(kt2))
+
+
+def test_gpu_deep_tree_correctness():
+    # n=2^22: DS=14 -> 7 global-scratch stack levels + 4 LDS + 2 register
+    # levels all exercised; reconstruction vs ground truth
+    n = 1 << 22
+    dpf = DPF(prf=DPF.PRF_AES128)
+    alphas = [0, n - 1, 123456, 3999999]
+    table = torch.randint(-(2**31), 2**31 - 1, (n, 16), dtype=torch.int64).to(
+        torch.int32
+    )
+    dpf.eval_init(table)
+    k1s, k2s = [], []
+    for a in alphas:
+        k1, k2 = dpf.gen(a, n)
+        k1s.append(k1)
+        k2s.append(k2)
+    ra = dpf.eval_gpu(k1s)
+    rb = dpf.eval_gpu(k2s)
+    rec = (ra.to(torch.int64) - rb.to(torch.int64)).to(torch.int32)
+    assert torch.equal(rec, table[alphas, :])
