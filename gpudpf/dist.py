@@ -28,7 +28,15 @@ class ShardedDPF(object):
     """Strong-scaling mode: one batch of keys is evaluated cooperatively by
     all ranks against a row-sharded table."""
 
-    def __init__(self, prf=None, device=None, group=None):
+    def __init__(self, prf=None, device=None, group=None,
+                 collective="all_reduce"):
+        """collective: "all_reduce" (default — the [batch,16] payload is
+        KB-scale and latency-bound, a fused all-reduce is optimal) or
+        "rs_ag" (explicit reduce-scatter of per-shard partials +
+        all-gather of results, NCCL/RCCL only — the decomposition a ring
+        all-reduce performs internally, exposed for batch-sharded
+        pipelines that consume only their own slice between the two
+        phases)."""
         if not td.is_initialized():
             raise Exception("torch.distributed must be initialized")
         self.group = group
@@ -37,6 +45,7 @@ class ShardedDPF(object):
         if self.world & (self.world - 1) != 0:
             raise Exception("world size must be a power of two")
         self.prf_method = DPF.DEFAULT_PRF if prf is None else prf
+        self.collective = collective
         self.local = DPF(prf=self.prf_method, device=device)
         self.table_num_entries = None
         self.table_effective_entry_size = None
@@ -91,8 +100,14 @@ class ShardedDPF(object):
         backend = td.get_backend(self.group)
         if backend == "nccl":
             dev = torch.device(self.local.device)
-            buf = part.to(dev)
-            td.all_reduce(buf, op=td.ReduceOp.SUM, group=self.group)
+            buf = part.to(dev).contiguous()
+            if self.collective == "rs_ag" and buf.shape[0] % self.world == 0:
+                chunk = torch.empty_like(buf[: buf.shape[0] // self.world])
+                td.reduce_scatter_tensor(chunk, buf, op=td.ReduceOp.SUM,
+                                         group=self.group)
+                td.all_gather_into_tensor(buf, chunk, group=self.group)
+            else:
+                td.all_reduce(buf, op=td.ReduceOp.SUM, group=self.group)
             return buf.cpu()
         buf = part.clone()
         td.all_reduce(buf, op=td.ReduceOp.SUM, group=self.group)
