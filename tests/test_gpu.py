@@ -263,3 +263,14 @@ def test_gpu_deep_tree_correctness():
     rb = dpf.eval_gpu(k2s)
     rec = (ra.to(torch.int64) - rb.to(torch.int64)).to(torch.int32)
     assert torch.equal(rec, table[alphas, :])
+
+
+def test_gpu_onehot_deep_tree():
+    # one-hot expansion kernel with global-scratch stack levels (n=2^20)
+    n = 1 << 20
+    dpf = DPF(prf=DPF.PRF_SALSA20)
+    k1, _ = dpf.gen(987654, n)
+    dpf.eval_init(torch.zeros((n, 1), dtype=torch.int32))
+    got = dpf.eval_gpu([k1], one_hot_only=True).numpy()[0]
+    want = _core.expand(k1.numpy(), DPF.PRF_SALSA20)
+    assert np.array_equal(got, want)
