@@ -168,6 +168,50 @@ __device__ __forceinline__ void salsa12_pair_core(uint4 seed, uint4& r0,
   r1 = make_uint4(y4 + seed.x, y3 + seed.y, y2 + seed.z, y1 + seed.w);
 }
 
+// Leaf-level variant: only the LOW output word (out[4] = x4 + seed.x) is
+// needed, so the final double-round computes just the dataflow cone of
+// x4: row-QR(x5,x6,x7,x4) needs post-column x5 (full QR), x6 (b,c,d of
+// its QR), x7 (b,c), x4 (b) — 39 ops instead of 96.
+#define SALSA_QR2_HEAD1(a, b, c, d, a2, b2, c2, d2) /* b only */ \
+  b ^= rotl(a + d, 7);   b2 ^= rotl(a2 + d2, 7)
+#define SALSA_QR2_HEAD2(a, b, c, d, a2, b2, c2, d2) /* b, c */ \
+  b ^= rotl(a + d, 7);   b2 ^= rotl(a2 + d2, 7);   \
+  c ^= rotl(b + a, 9);   c2 ^= rotl(b2 + a2, 9)
+#define SALSA_QR2_HEAD3(a, b, c, d, a2, b2, c2, d2) /* b, c, d */ \
+  b ^= rotl(a + d, 7);   b2 ^= rotl(a2 + d2, 7);   \
+  c ^= rotl(b + a, 9);   c2 ^= rotl(b2 + a2, 9);   \
+  d ^= rotl(c + b, 13);  d2 ^= rotl(c2 + b2, 13)
+
+__device__ __forceinline__ void salsa12_pair_low_core(uint4 seed, u32& lo0,
+                                                      u32& lo1) {
+  const u32 c0 = 0x65787061u, c5 = 0x6e642033u, c10 = 0x322d6279u,
+            c15 = 0x7465206bu;
+  u32 x0 = c0, x1 = seed.w, x2 = seed.z, x3 = seed.y, x4 = seed.x, x5 = c5,
+      x6 = 0, x7 = 0, x8 = 0, x9 = 0, x10 = c10, x11 = 0, x12 = 0, x13 = 0,
+      x14 = 0, x15 = c15;
+  u32 y0 = c0, y1 = x1, y2 = x2, y3 = x3, y4 = x4, y5 = c5, y6 = 0, y7 = 0,
+      y8 = 0, y9 = 1, y10 = c10, y11 = 0, y12 = 0, y13 = 0, y14 = 0, y15 = c15;
+#pragma unroll
+  for (int r = 0; r < 5; ++r) {
+    SALSA_QR2(x0, x4, x8, x12, y0, y4, y8, y12);
+    SALSA_QR2(x5, x9, x13, x1, y5, y9, y13, y1);
+    SALSA_QR2(x10, x14, x2, x6, y10, y14, y2, y6);
+    SALSA_QR2(x15, x3, x7, x11, y15, y3, y7, y11);
+    SALSA_QR2(x0, x1, x2, x3, y0, y1, y2, y3);
+    SALSA_QR2(x5, x6, x7, x4, y5, y6, y7, y4);
+    SALSA_QR2(x10, x11, x8, x9, y10, y11, y8, y9);
+    SALSA_QR2(x15, x12, x13, x14, y15, y12, y13, y14);
+  }
+  // 6th double-round, pruned to the cone of x4/y4:
+  SALSA_QR2_HEAD1(x0, x4, x8, x12, y0, y4, y8, y12);   // new x4 (b)
+  SALSA_QR2(x5, x9, x13, x1, y5, y9, y13, y1);         // new x5 (a: full)
+  SALSA_QR2_HEAD3(x10, x14, x2, x6, y10, y14, y2, y6); // new x6 (d)
+  SALSA_QR2_HEAD2(x15, x3, x7, x11, y15, y3, y7, y11); // new x7 (c)
+  SALSA_QR2_HEAD3(x5, x6, x7, x4, y5, y6, y7, y4);     // row: new x4 (d)
+  lo0 = x4 + seed.x;
+  lo1 = y4 + seed.x;
+}
+
 // ---------------------------------------------------------------------------
 // ChaCha20/12 (seed words 4..7 high->low, pos word 13, output words 4..7)
 // ---------------------------------------------------------------------------
@@ -228,6 +272,58 @@ __device__ __forceinline__ void chacha12_pair_core(uint4 seed, uint4& r0,
   }
   r0 = make_uint4(x7 + seed.x, x6 + seed.y, x5 + seed.z, x4 + seed.w);
   r1 = make_uint4(y7 + seed.x, y6 + seed.y, y5 + seed.z, y4 + seed.w);
+}
+
+// Leaf-level variant: only out[7] = x7 + seed.x is needed.  Final
+// double-round cone: diagonal QR(x2,x7,x8,x13) needs its full chain;
+// its inputs need x2 (full col QR), x7 (full: b needs a,c,d), x8 (col
+// QR c through the second c+=d), x13 (col QR d through the second
+// d-rotate) — 56 ops instead of 96.
+#define CHACHA_QR2_C10(a, b, c, d, a2, b2, c2, d2) /* c final (skip last b) */ \
+  a += b;  a2 += b2;  d ^= a;  d2 ^= a2;                            \
+  d = rotl(d, 16);  d2 = rotl(d2, 16);                              \
+  c += d;  c2 += d2;  b ^= c;  b2 ^= c2;                            \
+  b = rotl(b, 12);  b2 = rotl(b2, 12);                              \
+  a += b;  a2 += b2;  d ^= a;  d2 ^= a2;                            \
+  d = rotl(d, 8);  d2 = rotl(d2, 8);                                \
+  c += d;  c2 += d2
+#define CHACHA_QR2_D10(a, b, c, d, a2, b2, c2, d2) /* d final (skip last c,b) */ \
+  a += b;  a2 += b2;  d ^= a;  d2 ^= a2;                            \
+  d = rotl(d, 16);  d2 = rotl(d2, 16);                              \
+  c += d;  c2 += d2;  b ^= c;  b2 ^= c2;                            \
+  b = rotl(b, 12);  b2 = rotl(b2, 12);                              \
+  a += b;  a2 += b2;  d ^= a;  d2 ^= a2;                            \
+  d = rotl(d, 8);  d2 = rotl(d2, 8)
+
+__device__ __forceinline__ void chacha12_pair_low_core(uint4 seed, u32& lo0,
+                                                       u32& lo1) {
+  const u32 k0 = 0x65787061u, k1 = 0x6e642033u, k2 = 0x322d6279u,
+            k3 = 0x7465206bu;
+  u32 x0 = k0, x1 = k1, x2 = k2, x3 = k3;
+  u32 x4 = seed.w, x5 = seed.z, x6 = seed.y, x7 = seed.x;
+  u32 x8 = 0, x9 = 0, x10 = 0, x11 = 0, x12 = 0, x13 = 0, x14 = 0, x15 = 0;
+  u32 y0 = k0, y1 = k1, y2 = k2, y3 = k3;
+  u32 y4 = x4, y5 = x5, y6 = x6, y7 = x7;
+  u32 y8 = 0, y9 = 0, y10 = 0, y11 = 0, y12 = 0, y13 = 1, y14 = 0, y15 = 0;
+#pragma unroll
+  for (int r = 0; r < 5; ++r) {
+    CHACHA_QR2(x0, x4, x8, x12, y0, y4, y8, y12);
+    CHACHA_QR2(x1, x5, x9, x13, y1, y5, y9, y13);
+    CHACHA_QR2(x2, x6, x10, x14, y2, y6, y10, y14);
+    CHACHA_QR2(x3, x7, x11, x15, y3, y7, y11, y15);
+    CHACHA_QR2(x0, x5, x10, x15, y0, y5, y10, y15);
+    CHACHA_QR2(x1, x6, x11, x12, y1, y6, y11, y12);
+    CHACHA_QR2(x2, x7, x8, x13, y2, y7, y8, y13);
+    CHACHA_QR2(x3, x4, x9, x14, y3, y4, y9, y14);
+  }
+  // 6th double-round, pruned to the cone of x7/y7:
+  CHACHA_QR2_C10(x0, x4, x8, x12, y0, y4, y8, y12);   // x8 (c)
+  CHACHA_QR2_D10(x1, x5, x9, x13, y1, y5, y9, y13);   // x13 (d)
+  CHACHA_QR2(x2, x6, x10, x14, y2, y6, y10, y14);     // x2 (a: full)
+  CHACHA_QR2(x3, x7, x11, x15, y3, y7, y11, y15);     // x7 (b: full)
+  CHACHA_QR2(x2, x7, x8, x13, y2, y7, y8, y13);       // diagonal: x7 (b)
+  lo0 = x7 + seed.x;
+  lo1 = y7 + seed.x;
 }
 
 // ---------------------------------------------------------------------------
@@ -361,15 +457,9 @@ __device__ __forceinline__ void prf_pair_low(uint4 seed, const AesLds& T,
     r0 = seed.x * 4242u + 4242u;
     r1 = seed.x * 4243u + 4243u;
   } else if constexpr (PRF == PRF_SALSA20) {
-    uint4 a, b;
-    salsa12_pair_core(seed, a, b);
-    r0 = a.x;
-    r1 = b.x;
+    salsa12_pair_low_core(seed, r0, r1);
   } else if constexpr (PRF == PRF_CHACHA20) {
-    uint4 a, b;
-    chacha12_pair_core(seed, a, b);
-    r0 = a.x;
-    r1 = b.x;
+    chacha12_pair_low_core(seed, r0, r1);
   } else {
     uint4 a, b;
     aes_cipher_pair<true>(seed, T, a, b);
