@@ -33,13 +33,23 @@ def load_accuracy(dir_path):
     return out
 
 
+def _num(v):
+    try:
+        return float(v)
+    except (TypeError, ValueError):
+        return v
+
+
 def load_perf(csv_path):
+    """GPU kernel-sweep rows with numeric throughput and latency
+    (CPU-baseline rows in mixed CSVs are skipped)."""
     with open(csv_path) as f:
-        return [
-            {k: (float(v) if v.replace(".", "", 1).replace("-", "", 1).isdigit()
-                 else v) for k, v in row.items()}
-            for row in csv.DictReader(f)
-        ]
+        rows = [{k: _num(v) for k, v in row.items()}
+                for row in csv.DictReader(f)]
+    return [r for r in rows
+            if isinstance(r.get("throughput_dpfs_per_sec"), float)
+            and isinstance(r.get("latency_ms"), float)
+            and isinstance(r.get("num_entries"), float)]
 
 
 def join(acc_rows, perf_rows, max_latency_ms=100.0, max_comm_bytes=300_000,
