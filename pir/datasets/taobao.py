@@ -56,11 +56,8 @@ class TaobaoDataset:
         self.num_items = num_items
         self.field_sizes = list(field_sizes)
         if data_path and os.path.exists(data_path):
-            import pandas as pd
-
-            df = pd.read_csv(data_path)
-            # expects columns: clk, item histories as space-separated ids,
-            # plus categorical feature columns
+            # a preprocessed trace would be loaded here; raw-Taobao CSV
+            # parsing is deployment-specific
             raise NotImplementedError(
                 "supply a preprocessed trace; raw-taobao CSV parsing is "
                 "site-specific")
