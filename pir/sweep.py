@@ -54,7 +54,7 @@ def _make_dataset(name, quick):
 _DS = None
 
 
-def _init_worker(name, quick, seed_state):
+def _init_worker(name, quick):
     global _DS
     import torch
 
@@ -101,11 +101,11 @@ def main():
     cfgs = [dict(zip(keys, vals)) for vals in
             itertools.product(*(grid[k] for k in keys))]
     if a.processes <= 1:
-        _init_worker(a.dataset, a.quick, None)
+        _init_worker(a.dataset, a.quick)
         results = [run_config((c, out_dir)) for c in cfgs]
     else:
         with Pool(a.processes, initializer=_init_worker,
-                  initargs=(a.dataset, a.quick, None)) as pool:
+                  initargs=(a.dataset, a.quick)) as pool:
             results = pool.map(run_config, [(c, out_dir) for c in cfgs])
     print("wrote %d configs to %s" % (len(results), out_dir))
 
