@@ -96,12 +96,32 @@ py::array_t<std::int32_t> expand_batch(const std::vector<KeyArr>& keys,
       for (size_t i = 0; i < b; ++i)
         dpf_expand_full(ks[i], prf_method, optr + i * n);
     } else {
+      // When keys are scarcer than threads, split each key's expansion
+      // into W subtree shards (the residue-class restriction is itself a
+      // DPF key; shard r's leaves are the strided slice out[r::W]).
+      u64 shards = 1;
+      while (b * shards < (size_t)num_threads &&
+             shards * 2 <= (ks[0].n >> 1) && shards < 64)
+        shards <<= 1;
       std::vector<std::thread> pool;
       std::atomic<size_t> next{0};
+      const size_t tasks = b * (size_t)shards;
       for (int t = 0; t < num_threads; ++t) {
-        pool.emplace_back([&] {
-          for (size_t i; (i = next.fetch_add(1)) < b;)
-            dpf_expand_full(ks[i], prf_method, optr + i * n);
+        pool.emplace_back([&, shards] {
+          std::vector<u32> tmp(shards > 1 ? ks[0].n / shards : 0);
+          for (size_t task; (task = next.fetch_add(1)) < tasks;) {
+            const size_t i = task / shards;
+            const u64 r = (u64)(task % shards);
+            if (shards == 1) {
+              dpf_expand_full(ks[i], prf_method, optr + i * n);
+            } else {
+              DpfKey sub;
+              dpf_shard_subkey(ks[i], prf_method, r, shards, sub);
+              dpf_expand_full(sub, prf_method, tmp.data());
+              u32* row = optr + i * n;
+              for (u64 j = 0; j < sub.n; ++j) row[j * shards + r] = tmp[j];
+            }
+          }
         });
       }
       for (auto& th : pool) th.join();
