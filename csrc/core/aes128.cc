@@ -164,3 +164,70 @@ void aes128_encrypt_block(const unsigned char key[16], const unsigned char in[16
 }
 
 }  // namespace gpudpf
+
+// ---------------------------------------------------------------------------
+// AES-NI fast path (x86).  Bit-exact with the table implementation above
+// (both are FIPS-197 AES-128 on the same byte order); used by the hot CPU
+// expansion loop when the host supports it.
+// ---------------------------------------------------------------------------
+#if defined(__AES__) && defined(__SSE4_1__)
+#include <immintrin.h>
+
+namespace gpudpf {
+
+bool aesni_available() {
+  static const bool ok = __builtin_cpu_supports("aes");
+  return ok;
+}
+
+namespace {
+inline __m128i ks_step(__m128i key, __m128i kg) {
+  kg = _mm_shuffle_epi32(kg, _MM_SHUFFLE(3, 3, 3, 3));
+  key = _mm_xor_si128(key, _mm_slli_si128(key, 4));
+  key = _mm_xor_si128(key, _mm_slli_si128(key, 4));
+  key = _mm_xor_si128(key, _mm_slli_si128(key, 4));
+  return _mm_xor_si128(key, kg);
+}
+}  // namespace
+
+void aes128_expand_key_ni(const unsigned char key[16], AesNiRoundKeys& rk) {
+  __m128i* k = reinterpret_cast<__m128i*>(rk.rk);
+  k[0] = _mm_loadu_si128(reinterpret_cast<const __m128i*>(key));
+  k[1] = ks_step(k[0], _mm_aeskeygenassist_si128(k[0], 0x01));
+  k[2] = ks_step(k[1], _mm_aeskeygenassist_si128(k[1], 0x02));
+  k[3] = ks_step(k[2], _mm_aeskeygenassist_si128(k[2], 0x04));
+  k[4] = ks_step(k[3], _mm_aeskeygenassist_si128(k[3], 0x08));
+  k[5] = ks_step(k[4], _mm_aeskeygenassist_si128(k[4], 0x10));
+  k[6] = ks_step(k[5], _mm_aeskeygenassist_si128(k[5], 0x20));
+  k[7] = ks_step(k[6], _mm_aeskeygenassist_si128(k[6], 0x40));
+  k[8] = ks_step(k[7], _mm_aeskeygenassist_si128(k[7], 0x80));
+  k[9] = ks_step(k[8], _mm_aeskeygenassist_si128(k[8], 0x1b));
+  k[10] = ks_step(k[9], _mm_aeskeygenassist_si128(k[9], 0x36));
+}
+
+// Encrypt the two counter blocks (0 and 1) under one schedule, chains
+// interleaved.
+void aes128_encrypt2_ni(const AesNiRoundKeys& rk, unsigned char out0[16],
+                        unsigned char out1[16]) {
+  const __m128i* k = reinterpret_cast<const __m128i*>(rk.rk);
+  __m128i s0 = k[0];                                   // pt = 0
+  __m128i s1 = _mm_xor_si128(_mm_cvtsi32_si128(1), k[0]);  // pt = 1 (LE byte 0)
+  for (int r = 1; r < 10; ++r) {
+    s0 = _mm_aesenc_si128(s0, k[r]);
+    s1 = _mm_aesenc_si128(s1, k[r]);
+  }
+  s0 = _mm_aesenclast_si128(s0, k[10]);
+  s1 = _mm_aesenclast_si128(s1, k[10]);
+  _mm_storeu_si128(reinterpret_cast<__m128i*>(out0), s0);
+  _mm_storeu_si128(reinterpret_cast<__m128i*>(out1), s1);
+}
+
+}  // namespace gpudpf
+#else
+namespace gpudpf {
+bool aesni_available() { return false; }
+void aes128_expand_key_ni(const unsigned char[16], AesNiRoundKeys&) {}
+void aes128_encrypt2_ni(const AesNiRoundKeys&, unsigned char[16],
+                        unsigned char[16]) {}
+}  // namespace gpudpf
+#endif

@@ -127,7 +127,15 @@ namespace {
 inline void expand_children(const DpfKey& k, int prf_method, u128 seed, int i,
                             u128& c0, u128& c1) {
   u128 p0, p1;
-  if (prf_method == PRF_AES128) {
+  if (prf_method == PRF_AES128 && aesni_available()) {
+    unsigned char key[16], o0[16], o1[16];
+    std::memcpy(key, &seed, 16);
+    AesNiRoundKeys rk;
+    aes128_expand_key_ni(key, rk);
+    aes128_encrypt2_ni(rk, o0, o1);
+    std::memcpy(&p0, o0, 16);
+    std::memcpy(&p1, o1, 16);
+  } else if (prf_method == PRF_AES128) {
     unsigned char key[16], in0[16] = {0}, in1[16] = {0}, out[16];
     u32 rk[44];
     std::memcpy(key, &seed, 16);
@@ -154,28 +162,45 @@ inline void expand_children(const DpfKey& k, int prf_method, u128 seed, int i,
 // 2^ell), so leaves land in natural order directly.
 void dpf_expand_full(const DpfKey& k, int prf_method, u32* out) {
   const int depth = k.depth;
+  const bool vec8 = avx2_available() &&
+                    (prf_method == PRF_SALSA20 || prf_method == PRF_CHACHA20);
+  auto x8 = (prf_method == PRF_SALSA20) ? salsa12_x8 : chacha12_x8;
   std::vector<u128> cur(1, k.root), next;
+  u128 p0v[8], p1v[8];
   for (int l = 0; l < depth; ++l) {
     const int i = depth - 1 - l;  // eval-order level index
     const u64 width = (u64)1 << l;
-    if (i == 0) {
-      // Leaf level: emit low-32 shares straight into out.
-      for (u64 v = 0; v < width; ++v) {
-        u128 c0, c1;
-        expand_children(k, prf_method, cur[v], i, c0, c1);
+    const u64 w8 = vec8 ? (width & ~(u64)7) : 0;
+    if (i != 0) next.resize(width * 2);
+    // vectorized body: 8 parents per call, one call per child position
+    for (u64 v = 0; v < w8; v += 8) {
+      x8(&cur[v], 0, p0v);
+      x8(&cur[v], 1, p1v);
+      for (int j = 0; j < 8; ++j) {
+        const int sel = (int)(cur[v + j] & 1);
+        u128 c0 = p0v[j] + k.cw[sel][i * 2 + 0];
+        u128 c1 = p1v[j] + k.cw[sel][i * 2 + 1];
+        if (i == 0) {
+          out[v + j] = (u32)c0;
+          out[(v + j) | ((u64)1 << l)] = (u32)c1;
+        } else {
+          next[v + j] = c0;
+          next[(v + j) | ((u64)1 << l)] = c1;
+        }
+      }
+    }
+    for (u64 v = w8; v < width; ++v) {
+      u128 c0, c1;
+      expand_children(k, prf_method, cur[v], i, c0, c1);
+      if (i == 0) {
         out[v] = (u32)c0;
         out[v | ((u64)1 << l)] = (u32)c1;
-      }
-    } else {
-      next.resize(width * 2);
-      for (u64 v = 0; v < width; ++v) {
-        u128 c0, c1;
-        expand_children(k, prf_method, cur[v], i, c0, c1);
+      } else {
         next[v] = c0;
         next[v | ((u64)1 << l)] = c1;
       }
-      cur.swap(next);
     }
+    if (i != 0) cur.swap(next);
   }
   if (depth == 0) out[0] = (u32)k.root;
 }

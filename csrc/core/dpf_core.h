@@ -61,6 +61,20 @@ void aes128_encrypt_block_rk(const u32 rk[44], const unsigned char in[16],
 void aes128_tables(u32 te0[256], u32 te1[256], u32 te2[256], u32 te3[256],
                    u32 sbox[256]);
 
+// AES-NI fast path (x86 hosts; runtime-guarded, bit-exact with the table
+// implementation).  Used by the hot CPU expansion loop.
+struct AesNiRoundKeys {
+  alignas(16) unsigned char rk[11 * 16];
+};
+bool aesni_available();
+// AVX2 8-lane stream-cipher PRFs (runtime-guarded; prf_avx2.cc)
+bool avx2_available();
+void salsa12_x8(const u128* seeds, u32 pos, u128* out);
+void chacha12_x8(const u128* seeds, u32 pos, u128* out);
+void aes128_expand_key_ni(const unsigned char key[16], AesNiRoundKeys& rk);
+void aes128_encrypt2_ni(const AesNiRoundKeys& rk, unsigned char out0[16],
+                        unsigned char out1[16]);
+
 // ---------------------------------------------------------------------------
 // DPF key
 // ---------------------------------------------------------------------------

@@ -53,6 +53,7 @@ def build_core(force=False):
     sources = [
         os.path.join(CSRC, "core", "dpf_core.cc"),
         os.path.join(CSRC, "core", "prf.cc"),
+        os.path.join(CSRC, "core", "prf_avx2.cc"),
         os.path.join(CSRC, "core", "aes128.cc"),
         os.path.join(CSRC, "core", "core_bindings.cc"),
     ]
