@@ -27,25 +27,33 @@ QUICK_GRID = {
 }
 
 
-def _make_dataset(name, quick):
+def _make_dataset(name, size):
+    """size: quick (smoke), medium (full grid in minutes), full."""
     if name == "lm":
         from pir.datasets import language_model
 
-        kw = dict(vocab=512, corpus_len=20000) if quick else {}
+        kw = {"quick": dict(vocab=512, corpus_len=20000),
+              "medium": dict(vocab=1024, corpus_len=60000),
+              "full": {}}[size]
         ds = language_model.initialize(**kw)
-        ds.train_model(epochs=1, max_batches=20 if quick else None)
+        mb = {"quick": 20, "medium": 80, "full": None}[size]
+        ds.train_model(epochs=1, max_batches=mb)
     elif name == "movielens":
         from pir.datasets import movielens
 
-        kw = dict(num_items=512, num_users=400) if quick else {}
+        kw = {"quick": dict(num_items=512, num_users=400),
+              "medium": dict(num_items=2048, num_users=1500),
+              "full": {}}[size]
         ds = movielens.initialize(**kw)
-        ds.train_model(epochs=1 if quick else 2)
+        ds.train_model(epochs=1 if size == "quick" else 2)
     elif name == "taobao":
         from pir.datasets import taobao
 
-        kw = dict(num_items=512, num_samples=500) if quick else {}
+        kw = {"quick": dict(num_items=512, num_samples=500),
+              "medium": dict(num_items=4096, num_samples=2000),
+              "full": {}}[size]
         ds = taobao.initialize(**kw)
-        ds.train_model(epochs=1 if quick else 2)
+        ds.train_model(epochs=1 if size == "quick" else 2)
     else:
         raise ValueError(name)
     return ds
@@ -54,12 +62,12 @@ def _make_dataset(name, quick):
 _DS = None
 
 
-def _init_worker(name, quick):
+def _init_worker(name, size):
     global _DS
     import torch
 
     torch.manual_seed(0)
-    _DS = _make_dataset(name, quick)
+    _DS = _make_dataset(name, size)
 
 
 def run_config(args):
@@ -92,20 +100,23 @@ def main():
                     choices=["lm", "movielens", "taobao"])
     ap.add_argument("--out", default="pir/sweep_out")
     ap.add_argument("--quick", action="store_true")
+    ap.add_argument("--size", default=None,
+                    choices=["quick", "medium", "full"])
     ap.add_argument("--processes", type=int, default=8)
     a = ap.parse_args()
-    grid = QUICK_GRID if a.quick else GRID
+    size = a.size or ("quick" if a.quick else "full")
+    grid = QUICK_GRID if size == "quick" else GRID
     out_dir = os.path.join(a.out, a.dataset)
     os.makedirs(out_dir, exist_ok=True)
     keys = list(grid)
     cfgs = [dict(zip(keys, vals)) for vals in
             itertools.product(*(grid[k] for k in keys))]
     if a.processes <= 1:
-        _init_worker(a.dataset, a.quick)
+        _init_worker(a.dataset, size)
         results = [run_config((c, out_dir)) for c in cfgs]
     else:
         with Pool(a.processes, initializer=_init_worker,
-                  initargs=(a.dataset, a.quick)) as pool:
+                  initargs=(a.dataset, size)) as pool:
             results = pool.map(run_config, [(c, out_dir) for c in cfgs])
     print("wrote %d configs to %s" % (len(results), out_dir))
 
