@@ -10,12 +10,37 @@ namespace gpudpf {
 // RNG
 // ---------------------------------------------------------------------------
 KeyRng::KeyRng(const unsigned char* seed, size_t len) {
-  std::seed_seq seq(seed, seed + len);
-  gen_.seed(seq);
+  // Absorb the seed into a 128-bit chaining value with Davies-Meyer over
+  // AES-128 (h <- E_m(h) ^ h per 16-byte block, bit length appended), then
+  // expand the digest as the AES-128-CTR key.
+  unsigned char h[16] = {};
+  unsigned char blk[16];
+  auto absorb = [&h](const unsigned char m[16]) {
+    unsigned char e[16];
+    aes128_encrypt_block(m, h, e);
+    for (int i = 0; i < 16; ++i) h[i] ^= e[i];
+  };
+  for (size_t off = 0; off < len; off += 16) {
+    std::memset(blk, 0, sizeof(blk));
+    const size_t take = len - off < 16 ? len - off : 16;
+    std::memcpy(blk, seed + off, take);
+    absorb(blk);
+  }
+  std::memset(blk, 0, sizeof(blk));
+  const u64 bits = (u64)len * 8;
+  std::memcpy(blk, &bits, sizeof(bits));
+  absorb(blk);
+  aes128_expand_key(h, rk_);
 }
 
 u128 KeyRng::next_u128() {
-  return ((u128)gen_() << 64) | (u128)gen_();
+  unsigned char in[16], out[16];
+  const u128 c = ctr_++;
+  std::memcpy(in, &c, sizeof(in));
+  aes128_encrypt_block_rk(rk_, in, out);
+  u128 r;
+  std::memcpy(&r, out, sizeof(r));
+  return r;
 }
 
 u128 KeyRng::next_odd_u128() { return next_u128() | 1; }

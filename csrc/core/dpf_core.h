@@ -20,7 +20,7 @@
 #pragma once
 #include <cstdint>
 #include <cstring>
-#include <random>
+#include <cstddef>
 #include <stdexcept>
 #include <vector>
 
@@ -78,7 +78,11 @@ void aes128_encrypt2_ni(const AesNiRoundKeys& rk, unsigned char out0[16],
 // ---------------------------------------------------------------------------
 // DPF key
 // ---------------------------------------------------------------------------
-constexpr int kMaxDepth = 64;
+// The 524-int wire format carries 64 correction-word slots per selector =
+// 2*depth entries, so depth <= 32 (n <= 2^32) is a hard wire-format limit:
+// dpf_gen and key_deserialize both enforce it (a deeper key would serialize
+// silently truncated).
+constexpr int kMaxDepth = 32;
 constexpr int kKeyInts = 524;   // serialized size in int32 (2096 bytes)
 constexpr int kEntryWords = 16; // table entry = 16 x u32 (padded)
 
@@ -90,13 +94,21 @@ struct DpfKey {
 };
 
 // Seedable RNG for key generation (deterministic given seed bytes).
+// AES-128-CTR keyed by a Davies-Meyer digest of the seed bytes: unlike a
+// Mersenne Twister (which the reference uses, dpf_base/dpf.h RandGen —
+// and whose raw outputs, published verbatim as correction words, let one
+// server linearly reconstruct the generator state and recover the other
+// server's seeds), the published cw stream reveals nothing about the
+// generator key.  Callers pass os.urandom entropy for production keygen;
+// fixed seeds give deterministic keys for tests.
 class KeyRng {
  public:
   explicit KeyRng(const unsigned char* seed, size_t len);
   u128 next_u128();
   u128 next_odd_u128();
  private:
-  std::mt19937_64 gen_;
+  u32 rk_[44];   // AES-128 round keys (CTR mode)
+  u128 ctr_ = 0;
 };
 
 // Generate the two server keys for point alpha with payload beta over a

@@ -14,7 +14,13 @@ Public API parity (reference dpf.py:35-137):
     shares = d.eval_gpu([k1])   # [1, e] int32 secret shares
 """
 
-from gpudpf.dpf import DPF  # noqa: F401
-from gpudpf.dist import ReplicatedDPF, ShardedDPF  # noqa: F401
+try:
+    from gpudpf.dpf import DPF  # noqa: F401
+    from gpudpf.dist import ReplicatedDPF, ShardedDPF  # noqa: F401
+except ImportError:
+    # Fresh checkout: the native extensions are not built yet.  Importing
+    # the bare package must still work so `gpudpf._build` can bootstrap
+    # (`python -m gpudpf._build` or __graft_entry__.build()).
+    DPF = None  # type: ignore[assignment]
 
 __version__ = "0.1.0"

@@ -70,6 +70,30 @@ def test_deterministic_given_seed():
     assert not np.array_equal(a1[0], b1[0])
 
 
+def test_wire_format_depth_limit():
+    # The 524-int wire format holds 2*depth cw slots (64 max): depth <= 32
+    # is a hard limit and must raise, never silently truncate.
+    k1, k2 = _core.gen(7, 1 << 32, b"depth32", _core.PRF_DUMMY)  # max ok
+    assert int(k1[0]) == 32
+    with pytest.raises(Exception):
+        _core.gen(7, 1 << 33, b"depth33", _core.PRF_DUMMY)
+
+
+def test_keygen_csprng_stream():
+    # Keygen randomness is AES-CTR (not mt19937): correction words of keys
+    # from different seeds share no structure, and a one-bit seed change
+    # flips ~half the cw bits (sanity, not a statistical proof).
+    a = _core.gen(0, 4096, b"\x00" * 16, _core.PRF_DUMMY)[0]
+    b = _core.gen(0, 4096, b"\x01" + b"\x00" * 15, _core.PRF_DUMMY)[0]
+    cw_a = np.asarray(a[4:488], dtype=np.uint32)
+    cw_b = np.asarray(b[4:488], dtype=np.uint32)
+    used = cw_a != 0  # only populated slots
+    diff = np.unpackbits(
+        (cw_a[used] ^ cw_b[used]).view(np.uint8)
+    ).mean()
+    assert 0.4 < diff < 0.6
+
+
 def test_fused_cpu_oracle():
     n, e = 2048, 16
     alpha = 999
