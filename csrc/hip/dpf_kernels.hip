@@ -44,6 +44,7 @@
 #include <mutex>
 #include <stdexcept>
 #include <string>
+#include <vector>
 
 #include "dpf_hip_api.h"
 
@@ -857,23 +858,28 @@ size_t fused_shmem_bytes(int Z, int DS, int prf) {
 // fused/expand launches on DIFFERENT streams of one device would share
 // this buffer — the python API serializes launches per DPF call, which
 // is the supported pattern.
+//
+// Outgrown buffers are RETIRED, never freed: a hipGraph captured while a
+// launch used the old buffer holds its raw pointer and replays against it
+// (round-1 advisor finding — freeing here turned later replays into
+// use-after-free).  Growth is geometric (2x), so retired memory is bounded
+// by the live buffer's size and the retired list stays O(log largest).
 std::mutex g_scratch_mu;
 void* g_scratch[64] = {};
 size_t g_scratch_bytes[64] = {};
+std::vector<void*> g_scratch_retired[64];
 
-uint4* get_scratch(size_t bytes, hipStream_t stream) {
+uint4* get_scratch(size_t bytes, hipStream_t /*stream*/) {
   if (bytes == 0) return nullptr;
   int dev = 0;
   HIP_CHECK(hipGetDevice(&dev));
   std::lock_guard<std::mutex> lock(g_scratch_mu);
   if (g_scratch_bytes[dev] < bytes) {
-    if (g_scratch[dev]) {
-      HIP_CHECK(hipStreamSynchronize(stream));
-      HIP_CHECK(hipFree(g_scratch[dev]));
-      g_scratch[dev] = nullptr;
-    }
-    HIP_CHECK(hipMalloc(&g_scratch[dev], bytes));
-    g_scratch_bytes[dev] = bytes;
+    if (g_scratch[dev]) g_scratch_retired[dev].push_back(g_scratch[dev]);
+    size_t want = g_scratch_bytes[dev] ? g_scratch_bytes[dev] : bytes;
+    while (want < bytes) want *= 2;
+    HIP_CHECK(hipMalloc(&g_scratch[dev], want));
+    g_scratch_bytes[dev] = want;
   }
   return reinterpret_cast<uint4*>(g_scratch[dev]);
 }
