@@ -55,7 +55,7 @@ PRF_IDS = {
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--steps", type=int, default=50)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--entries", type=int, default=1 << 20)
     ap.add_argument("--prf", type=str, default="AES128", choices=list(PRF_IDS))
@@ -63,57 +63,101 @@ def main():
     ap.add_argument("--entry-size", type=int, default=16)
     ap.add_argument("--mode", type=str, default="replicate",
                     choices=["replicate", "shard"])
+    ap.add_argument("--device", type=str, default=None,
+                    help="cuda (default when available) or cpu (CI runs of "
+                         "the distributed path over gloo)")
+    ap.add_argument("--backend", type=str, default=None,
+                    choices=["nccl", "gloo"],
+                    help="process-group backend (default: nccl on cuda, "
+                         "gloo on cpu)")
+    ap.add_argument("--check", action="store_true",
+                    help="verify reconstruction correctness after timing "
+                         "(adds a second key batch eval)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     distributed = world > 1
+    on_gpu = (args.device or "cuda") != "cpu" and torch.cuda.is_available()
+    backend = args.backend or ("nccl" if on_gpu else "gloo")
     if distributed:
-        torch.distributed.init_process_group("nccl")
-        torch.cuda.set_device(local_rank)
-    device = "cuda:%d" % local_rank
+        torch.distributed.init_process_group(backend)
+        if on_gpu:
+            torch.cuda.set_device(local_rank)
+    device = ("cuda:%d" % local_rank) if on_gpu else "cpu"
     prf = PRF_IDS[args.prf]
     n = args.entries
 
-    # Synthetic workload: random-init table, random target indices.
-    torch.manual_seed(1000 + rank)
+    # Synthetic workload: random-init table, random target indices.  The
+    # table is rank-independent in shard mode (all ranks shard ONE table)
+    # and per-rank in replicate mode (independent serving streams).
+    tseed = 1000 if args.mode == "shard" else 1000 + rank
+    torch.manual_seed(tseed)
     table = torch.randint(-(2**31), 2**31 - 1, (n, args.entry_size),
                           dtype=torch.int32)
-    keys = []
-    g = torch.Generator().manual_seed(rank)
-    for i in range(args.batch):
-        alpha = int(torch.randint(0, n, (1,), generator=g).item())
-        k1, _ = _core.gen(alpha, n, b"bench-%d-%d" % (rank, i), prf)
+    # Keys: in shard mode every rank receives the SAME client batch (the
+    # world evaluates it cooperatively); in replicate mode each rank has
+    # its own stream.
+    kseed = 0 if args.mode == "shard" else rank
+    g = torch.Generator().manual_seed(kseed)
+    alphas = [int(torch.randint(0, n, (1,), generator=g).item())
+              for _ in range(args.batch)]
+    keys, keys2 = [], []
+    for i, alpha in enumerate(alphas):
+        k1, k2 = _core.gen(alpha, n, b"bench-%d-%d" % (kseed, i), prf)
         keys.append(torch.from_numpy(k1))
+        keys2.append(torch.from_numpy(k2))
     keys_cpu = torch.stack(keys).contiguous()
+    keys2_cpu = torch.stack(keys2).contiguous()
 
     if args.mode == "shard" and distributed:
         engine = ShardedDPF(prf=prf, device=device)
         engine.eval_init(table)
-        def step():
-            return engine.eval_gpu(keys_cpu)
+        if on_gpu:
+            def step(k=keys_cpu):
+                return engine.eval_gpu(k)
+        else:
+            def step(k=keys_cpu):
+                return engine.eval_cpu(k)
     else:
         engine = DPF(prf=prf, device=device)
         engine.eval_init(table)
-        def step():
-            return engine.eval_gpu(keys_cpu)
+        if on_gpu:
+            def step(k=keys_cpu):
+                return engine.eval_gpu(k)
+        else:
+            def step(k=keys_cpu):
+                return engine.eval_cpu(k)
+
+    def sync():
+        if on_gpu:
+            torch.cuda.synchronize()
 
     for _ in range(args.warmup):
         step()
-    torch.cuda.synchronize()
+    sync()
     if distributed:
         torch.distributed.barrier()
-        torch.cuda.synchronize()
+        sync()
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
         step()
-    torch.cuda.synchronize()
+    sync()
     t1 = time.perf_counter()
     if distributed:
         torch.distributed.barrier()
-        torch.cuda.synchronize()
+        sync()
+
+    if args.check:
+        a = step(keys_cpu).to(torch.int64)
+        b = step(keys2_cpu).to(torch.int64)
+        rec = (a - b).to(torch.int32)
+        want = table[alphas, : args.entry_size]
+        if not torch.equal(rec.cpu(), want):
+            raise SystemExit("bench --check FAILED: reconstruction mismatch "
+                             "(rank %d, mode %s)" % (rank, args.mode))
 
     elapsed = t1 - t0
     if distributed:
@@ -152,7 +196,8 @@ def main():
                 "batch_per_gpu": args.batch if args.mode == "replicate" else None,
                 "global_batch": args.batch * batches_per_step,
                 "parallelism": ("dp%d-replicated" % world) if args.mode == "replicate"
-                               else ("shard%d-rccl" % world),
+                               else ("shard%d-%s" % (world,
+                                     "rccl" if backend == "nccl" else backend)),
             },
         }), flush=True)
 

@@ -13,9 +13,16 @@ because truncation is a ring hom), so a single int32 sum all-reduce of the
 Payloads are KB-scale (batch x 64 B), so the collective is latency-bound:
 one fused all_reduce per batch, no bucketing needed.
 
+The GPU hot path is device-resident end to end: keys go host->device once,
+the fused kernel writes partials into a device buffer, the RCCL collective
+reduces that buffer in place over xGMI, and the result stays on device
+(`to_host=False`) for serving pipelines — zero host round-trips per batch.
+
 Also provided: ReplicatedDPF — each rank holds the full table and serves
 its own key stream (the weak-scaling / production throughput mode).
 """
+
+import warnings
 
 import torch
 import torch.distributed as td
@@ -72,7 +79,11 @@ class ShardedDPF(object):
         self.table_effective_entry_size = int(local_rows.shape[1])
         self.local.eval_init(local_rows.contiguous())
 
-    def _subkeys(self, keys):
+    def shard_subkeys(self, keys):
+        """Restrict a batch of full-domain wire-format keys to this rank's
+        residue class: [b, 524] int32 CPU tensor of depth-log2(W) subkeys.
+        Host-side (the restriction walks log2(W) PRF levels per key); on a
+        serving path do this once per batch, then feed eval_gpu_into."""
         if isinstance(keys, torch.Tensor) and keys.dim() == 2:
             keys = [keys[i] for i in range(keys.shape[0])]
         subs = [
@@ -85,32 +96,74 @@ class ShardedDPF(object):
         ]
         return torch.stack(subs)
 
-    def eval_gpu(self, keys):
-        part = self.local.eval_gpu(self._subkeys(keys))
-        part = self._allreduce(part)
-        return part
+    # backwards-compatible internal alias
+    _subkeys = shard_subkeys
+
+    def eval_gpu(self, keys, to_host=True):
+        """Evaluate a batch of full-domain keys cooperatively.  Partials
+        stay on device: fused kernel -> device buffer -> in-place RCCL
+        all-reduce over xGMI.  Returns [b, e] int32 shares — a CPU tensor
+        by default, or the device tensor when to_host=False (the serving
+        hot path does zero host round-trips)."""
+        subs = self.shard_subkeys(keys)
+        if self.local._entry_padded == DPF.ENTRY_SIZE:
+            # fused path, fully device-resident
+            dev = self.local._table_gpu.device
+            keys_gpu = subs.to(dev, non_blocking=True).contiguous()
+            part = torch.empty((subs.shape[0], DPF.ENTRY_SIZE),
+                               dtype=torch.int32, device=dev)
+            self.local.eval_gpu_into(keys_gpu, part)
+            part = self._allreduce_(part)
+            part = part[:, : self.table_effective_entry_size]
+        else:
+            # wide entries: two-stage MFMA path (device-resident output)
+            part = self.local.eval_gpu(subs, out_device=True)
+            part = self._allreduce_(part)
+        return part.cpu() if to_host else part
+
+    def eval_gpu_into(self, subkeys_gpu, out_gpu):
+        """Serving path: subkeys (from shard_subkeys) already on device,
+        fused partials written into out_gpu [b,16] and all-reduced in
+        place.  No host traffic at all."""
+        self.local.eval_gpu_into(subkeys_gpu, out_gpu)
+        self._allreduce_(out_gpu)
 
     def eval_cpu(self, keys):
-        part = self.local.eval_cpu(self._subkeys(keys))
-        part = self._allreduce(part)
-        return part
+        part = self.local.eval_cpu(self.shard_subkeys(keys))
+        return self._allreduce_(part.contiguous())
 
-    def _allreduce(self, part):
-        # int32 sum with wraparound == exact mod-2^32 reduction
+    def _allreduce_(self, buf):
+        """In-place int32 sum reduction of `buf` (wraparound == exact
+        mod-2^32).  For the nccl(=RCCL) backend buf must already live on
+        this rank's GPU — it is reduced in place with no host copies."""
         backend = td.get_backend(self.group)
         if backend == "nccl":
-            dev = torch.device(self.local.device)
-            buf = part.to(dev).contiguous()
-            if self.collective == "rs_ag" and buf.shape[0] % self.world == 0:
-                chunk = torch.empty_like(buf[: buf.shape[0] // self.world])
-                td.reduce_scatter_tensor(chunk, buf, op=td.ReduceOp.SUM,
-                                         group=self.group)
-                td.all_gather_into_tensor(buf, chunk, group=self.group)
+            if not buf.is_cuda:  # eval_cpu under an RCCL process group
+                dev = torch.device(self.local.device or "cuda")
+                gbuf = buf.to(dev)
+                td.all_reduce(gbuf, op=td.ReduceOp.SUM, group=self.group)
+                buf.copy_(gbuf.cpu())
+                return buf
+            if self.collective == "rs_ag":
+                if buf.shape[0] % self.world != 0:
+                    warnings.warn(
+                        "collective='rs_ag' needs batch %% world == 0 "
+                        "(batch=%d, world=%d); falling back to all_reduce "
+                        "for this batch" % (buf.shape[0], self.world))
+                    td.all_reduce(buf, op=td.ReduceOp.SUM, group=self.group)
+                else:
+                    chunk = torch.empty_like(buf[: buf.shape[0] // self.world])
+                    td.reduce_scatter_tensor(chunk, buf, op=td.ReduceOp.SUM,
+                                             group=self.group)
+                    td.all_gather_into_tensor(buf, chunk, group=self.group)
             else:
                 td.all_reduce(buf, op=td.ReduceOp.SUM, group=self.group)
-            return buf.cpu()
-        buf = part.clone()
-        td.all_reduce(buf, op=td.ReduceOp.SUM, group=self.group)
+            return buf
+        # gloo (CPU rendezvous / tests): collective runs on host memory
+        host = buf.cpu() if buf.is_cuda else buf
+        td.all_reduce(host, op=td.ReduceOp.SUM, group=self.group)
+        if buf.is_cuda:
+            buf.copy_(host)
         return buf
 
 

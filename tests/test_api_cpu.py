@@ -7,7 +7,7 @@ import numpy as np
 import pytest
 import torch
 
-from gpudpf import DPF
+from gpudpf import DPF, _core
 
 
 def test_cpu_dpf_one_hot():
@@ -134,3 +134,22 @@ def test_corrupt_key_rejected():
     garbage[0] = 99  # depth out of range
     with pytest.raises(Exception):
         dpf.eval_cpu([garbage], one_hot_only=True)
+
+
+def test_mixed_domain_batch_rejected():
+    # A batch mixing key domains must raise, not silently evaluate at the
+    # first key's depth (round-1 advisor finding).
+    d = DPF(prf=DPF.PRF_SALSA20, device="cpu")
+    ka, _ = d.gen(3, 1024)
+    kb, _ = d.gen(3, 2048)
+    with pytest.raises(Exception, match="domain"):
+        d.eval_cpu([ka, kb], one_hot_only=True)
+
+
+def test_large_domain_header_decode():
+    # n is a u64 in the wire header: a 2^32-entry domain (depth 32) must
+    # decode correctly, not truncate to the low 32 bits.
+    k1, _ = _core.gen(123, 1 << 32, b"big", _core.PRF_DUMMY)
+    d = DPF(prf=DPF.PRF_DUMMY, device="cpu")
+    kt, n, depth = d._keys_tensor(torch.from_numpy(k1).unsqueeze(0))
+    assert n == 1 << 32 and depth == 32

@@ -58,3 +58,67 @@ def test_sharded_gpu_eval_two_ranks():
         p.join(timeout=60)
     for rank, ok, err in results:
         assert ok, f"rank {rank}: {err}"
+
+
+def _nccl_worker(port, collective, q):
+    """world=1 RCCL process group on the single GPU: executes the real
+    nccl(=RCCL) collective branch of ShardedDPF._allreduce_ — device
+    all_reduce / reduce_scatter_tensor + all_gather_into_tensor — which a
+    gloo test can never reach.  On an 8-GPU node the identical code runs
+    with world=8 over xGMI."""
+    import torch.distributed as td
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    td.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        from gpudpf import DPF, ShardedDPF, _core
+
+        N = 1 << 14
+        prf = DPF.PRF_CHACHA20
+        torch.manual_seed(11)
+        table = torch.randint(-(2**31), 2**31 - 1, (N, 16),
+                              dtype=torch.int64).to(torch.int32)
+        idxs = [0, 999, N - 1]
+        ks = []
+        for i in idxs:
+            k1, k2 = _core.gen(i, N, b"nccl-%d" % i, prf)
+            ks.append((torch.from_numpy(k1), torch.from_numpy(k2)))
+
+        sd = ShardedDPF(prf=prf, device="cuda:0", collective=collective)
+        sd.eval_init(table)
+
+        # device-resident path: result must stay on the GPU
+        a_dev = sd.eval_gpu([k[0] for k in ks], to_host=False)
+        assert a_dev.is_cuda, "to_host=False must return a device tensor"
+        b_dev = sd.eval_gpu([k[1] for k in ks], to_host=False)
+        rec = (a_dev.to(torch.int64) - b_dev.to(torch.int64)).to(
+            torch.int32).cpu().numpy()
+        ok = bool(np.array_equal(rec, table[idxs, :].numpy()))
+
+        # serving path: subkeys pre-sharded, eval_gpu_into + in-place RCCL
+        subs1 = sd.shard_subkeys([k[0] for k in ks]).to("cuda:0").contiguous()
+        subs2 = sd.shard_subkeys([k[1] for k in ks]).to("cuda:0").contiguous()
+        o1 = torch.empty((len(idxs), 16), dtype=torch.int32, device="cuda:0")
+        o2 = torch.empty_like(o1)
+        sd.eval_gpu_into(subs1, o1)
+        sd.eval_gpu_into(subs2, o2)
+        rec2 = (o1.to(torch.int64) - o2.to(torch.int64)).to(
+            torch.int32).cpu().numpy()
+        ok = ok and bool(np.array_equal(rec2, table[idxs, :].numpy()))
+        q.put((0, ok, ""))
+    except Exception as e:  # pragma: no cover
+        q.put((0, False, repr(e)))
+    finally:
+        td.destroy_process_group()
+
+
+@pytest.mark.parametrize("collective", ["all_reduce", "rs_ag"])
+def test_sharded_rccl_branch_executes(collective):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    p = ctx.Process(target=_nccl_worker, args=(29797, collective, q))
+    p.start()
+    rank, ok, err = q.get(timeout=300)
+    p.join(timeout=60)
+    assert ok, f"rank {rank}: {err}"
