@@ -72,6 +72,10 @@ def test_leaf_perm_is_bijective_and_invertible(depth, idx_frac):
     prf=st.sampled_from(PRFS),
 )
 def test_prf_children_differ(seed_lo, seed_hi, prf):
+    if prf == _core.PRF_DUMMY and seed_lo == seed_hi == 2**64 - 1:
+        # DUMMY is (seed+1)*(pos+4242) mod 2^128: the all-ones seed is its
+        # one degenerate fixed point (both children 0).  Test-only PRF.
+        return
     r0 = _core.prf(prf, seed_lo, seed_hi, 0)
     r1 = _core.prf(prf, seed_lo, seed_hi, 1)
     assert r0 != r1  # (holds for all four PRFs over random seeds)
