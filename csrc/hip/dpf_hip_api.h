@@ -30,6 +30,18 @@ void launch_naive(std::uintptr_t keys, std::uintptr_t out,
 void launch_prf_sol(std::uintptr_t aes_tabs, std::uintptr_t out, int blocks,
                     int iters, int prf, std::uintptr_t stream);
 
+// Unit-test probes (tests/test_gpu_alu.py; analog of the reference's
+// dpf_gpu/tests/test_128_bit.cu).  All pointers are device int32/uint32
+// buffers; u128 values are 4xu32 limbs little-endian.
+void launch_probe_alu(std::uintptr_t a, std::uintptr_t b,
+                      std::uintptr_t add_out, std::uintptr_t mul_out,
+                      int count, std::uintptr_t stream);
+void launch_probe_prf(std::uintptr_t seeds, std::uintptr_t aes_tabs,
+                      std::uintptr_t pair0, std::uintptr_t pair1,
+                      std::uintptr_t single0, std::uintptr_t single1,
+                      std::uintptr_t low0, std::uintptr_t low1, int count,
+                      int prf, std::uintptr_t stream);
+
 // Exact u128 GEMM (research harness; gemm128.hip).  a: [M,K] u128 (as
 // 4xint32 limbs), bt: [N,K] u128, c: [M,N] u128, partials: device scratch
 // of gemm128_ksplit(M,N,K)*M*N u128.

@@ -839,6 +839,118 @@ void launch_prf_sol(std::uintptr_t aes_tabs, std::uintptr_t out, int blocks,
 }
 
 // ---------------------------------------------------------------------------
+// ALU / PRF probe kernels — the unit-test analog of the reference's
+// dpf_gpu/tests/test_128_bit.cu:192-200 (device add/mul/pack asserted
+// against host __int128): one element per thread, results compared
+// elementwise against the CPU core in tests/test_gpu_alu.py.  These probe
+// the EXACT device functions the production kernels inline (add128,
+// prf_pair, prf_pair_low, prf_full with the replicated-LDS AES table), so
+// a PRF-core bug fails a unit test, not only end-to-end reconstruction.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void probe_alu_kernel(
+    const uint4* __restrict__ a, const uint4* __restrict__ b,
+    uint4* __restrict__ add_out, uint4* __restrict__ mul_out, int count) {
+  const int i = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (i >= count) return;
+  add_out[i] = add128(a[i], b[i]);
+  const unsigned __int128 x =
+      ((unsigned __int128)(((u64)a[i].w << 32) | a[i].z) << 64) |
+      (((u64)a[i].y << 32) | a[i].x);
+  const unsigned __int128 y =
+      ((unsigned __int128)(((u64)b[i].w << 32) | b[i].z) << 64) |
+      (((u64)b[i].y << 32) | b[i].x);
+  const unsigned __int128 r = x * y;
+  const u64 lo = (u64)r, hi = (u64)(r >> 64);
+  mul_out[i] = make_uint4((u32)lo, (u32)(lo >> 32), (u32)hi, (u32)(hi >> 32));
+}
+
+template <int PRF>
+__global__ __launch_bounds__(256) void probe_prf_kernel(
+    const uint4* __restrict__ seeds, const u32* __restrict__ aes_tabs,
+    uint4* __restrict__ pair0, uint4* __restrict__ pair1,
+    uint4* __restrict__ single0, uint4* __restrict__ single1,
+    u32* __restrict__ low0, u32* __restrict__ low1, int count) {
+  extern __shared__ u32 smem[];
+  u32* aes_lds = smem;
+  if constexpr (PRF == PRF_AES128) {
+    for (int e = (int)threadIdx.x; e < 256; e += blockDim.x) {
+      const u32 v = aes_tabs[e];
+#pragma unroll
+      for (int c = 0; c < AES_REP; ++c) aes_lds[e * AES_REP + c] = v;
+    }
+    __syncthreads();
+  }
+  AesLds T{aes_lds, (u32)(threadIdx.x & (AES_REP - 1))};
+  const int i = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (i >= count) return;
+  const uint4 s = seeds[i];
+  uint4 r0, r1;
+  prf_pair<PRF>(s, T, r0, r1);
+  pair0[i] = r0;
+  pair1[i] = r1;
+  single0[i] = prf_full<PRF>(s, 0, T);
+  single1[i] = prf_full<PRF>(s, 1, T);
+  u32 l0, l1;
+  prf_pair_low<PRF>(s, T, l0, l1);
+  low0[i] = l0;
+  low1[i] = l1;
+}
+
+void launch_probe_alu(std::uintptr_t a, std::uintptr_t b,
+                      std::uintptr_t add_out, std::uintptr_t mul_out,
+                      int count, std::uintptr_t stream) {
+  auto st = reinterpret_cast<hipStream_t>(stream);
+  const int blocks = (count + 255) / 256;
+  hipLaunchKernelGGL(probe_alu_kernel, dim3(blocks), dim3(256), 0, st,
+                     reinterpret_cast<const uint4*>(a),
+                     reinterpret_cast<const uint4*>(b),
+                     reinterpret_cast<uint4*>(add_out),
+                     reinterpret_cast<uint4*>(mul_out), count);
+  HIP_CHECK(hipGetLastError());
+}
+
+void launch_probe_prf(std::uintptr_t seeds, std::uintptr_t aes_tabs,
+                      std::uintptr_t pair0, std::uintptr_t pair1,
+                      std::uintptr_t single0, std::uintptr_t single1,
+                      std::uintptr_t low0, std::uintptr_t low1, int count,
+                      int prf, std::uintptr_t stream) {
+  auto st = reinterpret_cast<hipStream_t>(stream);
+  const int blocks = (count + 255) / 256;
+  const size_t shmem = (prf == PRF_AES128) ? AES_LDS_WORDS * 4 : 0;
+  auto* sp = reinterpret_cast<const uint4*>(seeds);
+  auto* ap = reinterpret_cast<const u32*>(aes_tabs);
+  auto* p0 = reinterpret_cast<uint4*>(pair0);
+  auto* p1 = reinterpret_cast<uint4*>(pair1);
+  auto* s0 = reinterpret_cast<uint4*>(single0);
+  auto* s1 = reinterpret_cast<uint4*>(single1);
+  auto* l0 = reinterpret_cast<u32*>(low0);
+  auto* l1 = reinterpret_cast<u32*>(low1);
+  switch (prf) {
+    case PRF_DUMMY:
+      hipLaunchKernelGGL(probe_prf_kernel<PRF_DUMMY>, dim3(blocks), dim3(256),
+                         shmem, st, sp, ap, p0, p1, s0, s1, l0, l1, count);
+      break;
+    case PRF_SALSA20:
+      hipLaunchKernelGGL(probe_prf_kernel<PRF_SALSA20>, dim3(blocks),
+                         dim3(256), shmem, st, sp, ap, p0, p1, s0, s1, l0, l1,
+                         count);
+      break;
+    case PRF_CHACHA20:
+      hipLaunchKernelGGL(probe_prf_kernel<PRF_CHACHA20>, dim3(blocks),
+                         dim3(256), shmem, st, sp, ap, p0, p1, s0, s1, l0, l1,
+                         count);
+      break;
+    case PRF_AES128:
+      hipLaunchKernelGGL(probe_prf_kernel<PRF_AES128>, dim3(blocks), dim3(256),
+                         shmem, st, sp, ap, p0, p1, s0, s1, l0, l1, count);
+      break;
+    default:
+      throw std::invalid_argument("unknown PRF");
+  }
+  HIP_CHECK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
 // Host launchers
 // ---------------------------------------------------------------------------
 namespace {

@@ -85,6 +85,14 @@ PYBIND11_MODULE(_hip, m) {
   m.def("gemm_u32_mfma", &gpudpf_hip::launch_gemm_u32_mfma, py::arg("da"),
         py::arg("dbt"), py::arg("c"), py::arg("m"), py::arg("n"), py::arg("k"),
         py::arg("stream"), py::call_guard<py::gil_scoped_release>());
+  m.def("probe_alu", &gpudpf_hip::launch_probe_alu, py::arg("a"), py::arg("b"),
+        py::arg("add_out"), py::arg("mul_out"), py::arg("count"),
+        py::arg("stream"), py::call_guard<py::gil_scoped_release>());
+  m.def("probe_prf", &gpudpf_hip::launch_probe_prf, py::arg("seeds"),
+        py::arg("aes_tabs"), py::arg("pair0"), py::arg("pair1"),
+        py::arg("single0"), py::arg("single1"), py::arg("low0"),
+        py::arg("low1"), py::arg("count"), py::arg("prf"), py::arg("stream"),
+        py::call_guard<py::gil_scoped_release>());
   m.def("ensure_aes_tables", &ensure_aes_tables, py::arg("device"));
   m.def("device_count", &device_count);
 }
