@@ -169,6 +169,28 @@ py::array_t<std::int64_t> leaf_perm_table(u64 n, int zlog) {
   return out;
 }
 
+// Permuted rows for a SUBSET of natural indices (leaf_perm is a pure bit
+// permutation, so it never needs the full n-entry map): lets eval_init
+// stream multi-hundred-GB tables chunk by chunk without materializing an
+// 8-byte-per-row permutation table (34 GB at n=2^32).
+py::array_t<std::int64_t> leaf_perm_rows(
+    py::array_t<std::int64_t, py::array::c_style | py::array::forcecast> idx,
+    u64 n, int zlog) {
+  const auto m = idx.size();
+  py::array_t<std::int64_t> out(m);
+  auto* p = out.mutable_data();
+  const auto* q = idx.data();
+  {
+    py::gil_scoped_release nogil;
+    for (py::ssize_t i = 0; i < m; ++i) {
+      const u64 v = (u64)q[i];
+      if (v >= n) throw std::out_of_range("index out of range");
+      p[i] = (std::int64_t)leaf_perm(n, zlog, v);
+    }
+  }
+  return out;
+}
+
 py::tuple prf(int method, u64 seed_lo, u64 seed_hi, u64 pos) {
   u128 s = ((u128)seed_hi << 64) | seed_lo;
   u128 r = prf_eval(method, s, (u128)pos);
@@ -288,6 +310,7 @@ PYBIND11_MODULE(_core, m) {
   m.def("eval_point128", &eval_point128);
   m.def("eval_fused_cpu", &eval_fused_cpu);
   m.def("leaf_perm_table", &leaf_perm_table);
+  m.def("leaf_perm_rows", &leaf_perm_rows);
   m.def("zlog_for_depth", &zlog_for_depth);
   m.def("prf", &prf);
   m.def("aes_block", &aes_block);

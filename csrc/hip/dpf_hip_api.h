@@ -59,4 +59,12 @@ void launch_gemm_u32_mfma(std::uintptr_t da, std::uintptr_t dbt,
                           std::uintptr_t c, long long M, long long N,
                           long long K, std::uintptr_t stream);
 
+// Streaming mod-2^32 GEMM (gemm_stream.hip): reads the u32 table in place
+// (no transposed copy, no digit planes) — the huge-table wide-entry path.
+// a: [batch, K] u32 shares; b: [K, N] u32 table; c: [batch, N] u32
+// (zeroed by caller).  batch <= 64 (python wrapper chunks larger).
+void launch_gemm_u32_stream(std::uintptr_t a, std::uintptr_t b,
+                            std::uintptr_t c, int batch, long long K,
+                            long long N, std::uintptr_t stream);
+
 }  // namespace gpudpf_hip

@@ -85,6 +85,10 @@ PYBIND11_MODULE(_hip, m) {
   m.def("gemm_u32_mfma", &gpudpf_hip::launch_gemm_u32_mfma, py::arg("da"),
         py::arg("dbt"), py::arg("c"), py::arg("m"), py::arg("n"), py::arg("k"),
         py::arg("stream"), py::call_guard<py::gil_scoped_release>());
+  m.def("gemm_u32_stream", &gpudpf_hip::launch_gemm_u32_stream, py::arg("a"),
+        py::arg("b"), py::arg("c"), py::arg("batch"), py::arg("k"),
+        py::arg("n"), py::arg("stream"),
+        py::call_guard<py::gil_scoped_release>());
   m.def("probe_alu", &gpudpf_hip::launch_probe_alu, py::arg("a"), py::arg("b"),
         py::arg("add_out"), py::arg("mul_out"), py::arg("count"),
         py::arg("stream"), py::call_guard<py::gil_scoped_release>());

@@ -76,6 +76,7 @@ def build_hip(force=False):
         os.path.join(CSRC, "hip", "dpf_kernels.hip"),
         os.path.join(CSRC, "hip", "gemm128.hip"),
         os.path.join(CSRC, "hip", "gemm_u32.hip"),
+        os.path.join(CSRC, "hip", "gemm_stream.hip"),
         os.path.join(CSRC, "hip", "hip_bindings.cc"),
         os.path.join(CSRC, "core", "aes128.cc"),
     ]

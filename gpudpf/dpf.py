@@ -96,47 +96,123 @@ class DPF(object):
     # ------------------------------------------------------------------
     # Server side
     # ------------------------------------------------------------------
-    def eval_init(self, table):
-        """Upload an [n, e] int32 table.  n is padded to the next
-        power-of-two domain (>= 128) and e to a multiple of 16; rows are
-        reordered by the kernel layout contract (leaf_perm).  Entries
-        wider than 16 words are served by the two-stage MFMA path."""
-        self.table = table
-        self.table_num_entries = int(table.shape[0])
-        self.table_effective_entry_size = int(table.shape[1])
-        n, e = self.table_num_entries, self.table_effective_entry_size
+    # Largest domain for which the full natural->row permutation tensor is
+    # kept on device (needed only by the one-hot output path's un-permute
+    # gather; 2^27 rows = 1 GiB of int64).  Larger domains compute row
+    # permutations chunk-wise on the host (leaf_perm is a bit permutation).
+    PERM_MATERIALIZE_MAX = 1 << 27
+    # eval_init upload granularity (bytes of table rows per H2D chunk)
+    INIT_CHUNK_BYTES = 256 << 20
 
+    def _setup_domain(self, n, e):
+        """Common eval_init metadata: domain padding, kernel layout
+        parameters, device resolution."""
+        self.table_num_entries = int(n)
+        self.table_effective_entry_size = int(e)
         # extensions over the reference: non-power-of-two n (zero-padded to
         # the next power-of-two domain) and entries wider than 16 words
-        # (served by the two-stage MFMA path; reference TODOs dpf.py:16-24)
+        # (served by the two-stage GEMM paths; reference TODOs dpf.py:16-24)
         nd = self._domain(n)
         ep = -(-e // self.ENTRY_SIZE) * self.ENTRY_SIZE
         self._n_domain = nd
         self._entry_padded = ep
-
         self._depth = nd.bit_length() - 1
         self._zlog = _core.zlog_for_depth(self._depth)
-
         if self.device is None:
             self.device = "cuda:0" if torch.cuda.is_available() else "cpu"
-        dev = torch.device(self.device)
+        return torch.device(self.device)
 
-        perm = torch.from_numpy(_core.leaf_perm_table(nd, self._zlog))
-        if dev.type == "cuda":
-            # pad + permute on the GPU: the row scatter of a multi-GB
-            # table is memory-bandwidth work, not host work
-            padded = torch.zeros((nd, ep), dtype=torch.int32, device=dev)
-            padded[:n, :e] = table.to(torch.int32).to(dev, non_blocking=True)
-            self._perm_gpu = perm.to(dev)
-            reordered = torch.empty_like(padded)
-            reordered[self._perm_gpu] = padded  # row perm[i] <- natural row i
-            self._table_gpu = reordered.contiguous()
-            del padded
-            if self.prf_method == self.PRF_AES128:
-                self._aes_ptr = _hip.ensure_aes_tables(dev.index or 0)
-        else:
+    def _perm_rows(self, indices):
+        """Permuted table rows for a tensor of natural indices (host
+        compute; works for any domain size without the full perm map)."""
+        return torch.from_numpy(
+            _core.leaf_perm_rows(indices.to(torch.int64).cpu().numpy(),
+                                 self._n_domain, self._zlog))
+
+    def eval_init(self, table):
+        """Upload an [n, e] int32 table.  n is padded to the next
+        power-of-two domain (>= 128) and e to a multiple of 16; rows are
+        reordered by the kernel layout contract (leaf_perm).  Entries
+        wider than 16 words are served by the two-stage GEMM paths.
+
+        The upload streams in row chunks: peak device memory is the
+        padded table plus one chunk (the round-1 implementation built a
+        second full-size padded copy first, capping tables at half of
+        HBM)."""
+        self.table = table
+        n, e = int(table.shape[0]), int(table.shape[1])
+        dev = self._setup_domain(n, e)
+        if dev.type != "cuda":
             self._table_gpu = None
             self._perm_gpu = None
+            return
+        nd, ep = self._n_domain, self._entry_padded
+        self._table_gpu = torch.zeros((nd, ep), dtype=torch.int32, device=dev)
+        if nd <= self.PERM_MATERIALIZE_MAX:
+            self._perm_gpu = torch.from_numpy(
+                _core.leaf_perm_table(nd, self._zlog)).to(dev)
+        else:
+            self._perm_gpu = None
+        t32 = table.to(torch.int32)
+        rows_per_chunk = max(1, self.INIT_CHUNK_BYTES // (ep * 4))
+        for lo in range(0, n, rows_per_chunk):
+            hi = min(n, lo + rows_per_chunk)
+            if self._perm_gpu is not None:
+                rows = self._perm_gpu[lo:hi]
+            else:
+                rows = self._perm_rows(torch.arange(lo, hi)).to(dev)
+            self._table_gpu[rows, :e] = t32[lo:hi].to(dev, non_blocking=False)
+        if self.prf_method == self.PRF_AES128:
+            self._aes_ptr = _hip.ensure_aes_tables(dev.index or 0)
+
+    def eval_init_empty(self, n, e):
+        """Allocate a zeroed device table for an [n, e] domain WITHOUT host
+        table data — the huge-table path (tables near HBM capacity are
+        never materialized on the host; fill them with table_write).  The
+        CPU fallback paths (eval_cpu with a table) are unavailable."""
+        self.table = None
+        dev = self._setup_domain(n, e)
+        if dev.type != "cuda":
+            raise Exception("eval_init_empty requires a GPU device")
+        nd, ep = self._n_domain, self._entry_padded
+        self._table_gpu = torch.zeros((nd, ep), dtype=torch.int32, device=dev)
+        if nd <= self.PERM_MATERIALIZE_MAX:
+            self._perm_gpu = torch.from_numpy(
+                _core.leaf_perm_table(nd, self._zlog)).to(dev)
+        else:
+            self._perm_gpu = None
+        if self.prf_method == self.PRF_AES128:
+            self._aes_ptr = _hip.ensure_aes_tables(dev.index or 0)
+
+    def table_write(self, indices, rows):
+        """Streaming ingest: write natural-order rows [m, e] int32 at
+        natural indices [m] into the (permuted) device table."""
+        if self._table_gpu is None:
+            raise Exception("call eval_init/eval_init_empty first")
+        e = self.table_effective_entry_size
+        if rows.dim() != 2 or int(rows.shape[1]) != e:
+            raise Exception("rows must be [m, %d]" % e)
+        dev = self._table_gpu.device
+        idx = indices if isinstance(indices, torch.Tensor) \
+            else torch.as_tensor(indices, dtype=torch.int64)
+        if self._perm_gpu is not None:
+            prows = self._perm_gpu[idx.to(dev)]
+        else:
+            prows = self._perm_rows(idx).to(dev)
+        self._table_gpu[prows, :e] = rows.to(torch.int32).to(dev)
+
+    def table_read(self, indices):
+        """Gather natural-order rows [m, e] int32 from the device table."""
+        if self._table_gpu is None:
+            raise Exception("call eval_init/eval_init_empty first")
+        dev = self._table_gpu.device
+        idx = indices if isinstance(indices, torch.Tensor) \
+            else torch.as_tensor(indices, dtype=torch.int64)
+        if self._perm_gpu is not None:
+            prows = self._perm_gpu[idx.to(dev)]
+        else:
+            prows = self._perm_rows(idx).to(dev)
+        return self._table_gpu[prows, : self.table_effective_entry_size]
 
     def eval_free(self):
         self._table_gpu = None
@@ -201,6 +277,10 @@ class DPF(object):
             chunk = keys_gpu[lo:hi].contiguous()
             b = hi - lo
             if one_hot_only:
+                if self._perm_gpu is None:
+                    raise Exception(
+                        "one_hot_only needs the materialized row "
+                        "permutation (domain > PERM_MATERIALIZE_MAX)")
                 out = torch.empty((b, n), dtype=torch.int32, device=dev)
                 _hip.eval_expand(
                     chunk.data_ptr(), out.data_ptr(), self._aes_ptr, b, n,
@@ -260,7 +340,17 @@ class DPF(object):
             s_matmul.wait_stream(s_expand)
             shares.record_stream(s_matmul)
             with torch.cuda.stream(s_matmul):
-                outs.append(ops.pir_matmul_u32(shares, self._table_gpu))
+                # GEMM dispatch: the MFMA digit-plane path materializes a
+                # transposed copy + int8 planes of the table (2x its
+                # bytes) — a win only when the table is small enough to
+                # afford that and the batch is large enough to be
+                # compute-bound.  Otherwise stream the u32 table in place.
+                table_bytes = self._table_gpu.numel() * 4
+                if table_bytes <= (2 << 30) and (hi - lo) >= 128:
+                    outs.append(ops.pir_matmul_u32(shares, self._table_gpu))
+                else:
+                    outs.append(
+                        ops.pir_matmul_u32_stream(shares, self._table_gpu))
         torch.cuda.current_stream(dev).wait_stream(s_matmul)
         torch.cuda.current_stream(dev).wait_stream(s_expand)
         out = torch.cat(outs) if len(outs) > 1 else outs[0]

@@ -53,6 +53,34 @@ def _digit_planes(x_u32_gpu):
     return out
 
 
+def pir_matmul_u32_stream(shares, table, out=None):
+    """C = shares @ table mod 2^32, streaming the u32 table in place.
+
+    shares: [M, K] int32, table: [K, N] int32 (both already on the GPU for
+    the huge-table path).  Unlike pir_matmul_u32 this allocates NO copy of
+    the table (no transpose, no digit planes) — the only extra memory is
+    the [M, N] output — so it serves tables sized to HBM capacity
+    (200+ GB).  Batches > 64 are evaluated in 64-row chunks, each chunk
+    re-streaming the table once."""
+    assert shares.dtype == torch.int32 and table.dtype == torch.int32
+    M, K = shares.shape
+    K2, N = table.shape
+    assert K == K2
+    dev = torch.device("cuda:0") if shares.device.type == "cpu" else shares.device
+    a = shares.to(dev).contiguous()
+    b = table.to(dev).contiguous()
+    c = out if out is not None else torch.empty(
+        (M, N), dtype=torch.int32, device=dev)
+    stream = torch.cuda.current_stream(dev).cuda_stream
+    for lo in range(0, M, 64):
+        hi = min(M, lo + 64)
+        chunk = c[lo:hi]
+        chunk.zero_()
+        _hip.gemm_u32_stream(a[lo:hi].data_ptr(), b.data_ptr(),
+                             chunk.data_ptr(), hi - lo, K, N, stream)
+    return c
+
+
 def pir_matmul_u32(shares, table):
     """C = shares @ table mod 2^32 on the MFMA matrix cores.
 
