@@ -839,6 +839,78 @@ void launch_prf_sol(std::uintptr_t aes_tabs, std::uintptr_t out, int blocks,
 }
 
 // ---------------------------------------------------------------------------
+// Level-synchronized breadth-first expansion (the reference's
+// dpf_breadth_first.cu:35-103 strategy, completing the strategy matrix
+// next to naive / fused / two-stage).  One kernel launch per tree level;
+// ping-pong u128 seed frontiers in global scratch; the last level writes
+// low-32 one-hot shares in NATURAL index order with two coalesced
+// streams (leaf idx = parent_pos | bit << (depth-1), because the eval
+// recurrence consumes index bits LSB-first).
+//
+// This strategy pays n*16 B of frontier traffic per level pair and
+// depth kernel launches — the fused DFS kernel exists precisely to avoid
+// that; BFS is kept as a measured research point, not the production
+// path.
+// ---------------------------------------------------------------------------
+template <int PRF>
+__global__ __launch_bounds__(256) void dpf_bfs_level_kernel(
+    const int* __restrict__ keys, const uint4* __restrict__ parents,
+    uint4* __restrict__ children, u32* __restrict__ out,
+    const u32* __restrict__ aes_tabs, int level, int depth, long long n) {
+  extern __shared__ u32 smem[];
+  u32* aes_lds = smem;
+  uint4* cw_lvl = reinterpret_cast<uint4*>(
+      smem + (PRF == PRF_AES128 ? AES_LDS_WORDS : 0));
+  const int t = (int)threadIdx.x;
+  const int key_id = (int)blockIdx.y;
+  const long long key_base = (long long)key_id * 524;
+  if constexpr (PRF == PRF_AES128) {
+    for (int e = t; e < 256; e += blockDim.x) {
+      const u32 v = aes_tabs[e];
+#pragma unroll
+      for (int c = 0; c < AES_REP; ++c) aes_lds[e * AES_REP + c] = v;
+    }
+  }
+  // stage this level's 4 correction words (cw[sel][2i+b], i = eval level)
+  const int i_eval = depth - 1 - level;
+  if (t < 4) {
+    const int sel = t >> 1, b = t & 1;
+    cw_lvl[t] = reinterpret_cast<const uint4*>(keys + key_base + 4)
+        [sel * 64 + i_eval * 2 + b];
+  }
+  __syncthreads();
+  AesLds T{aes_lds, (u32)(t & (AES_REP - 1))};
+
+  const long long n_parents = (long long)1 << level;
+  const long long p = (long long)blockIdx.x * blockDim.x + t;
+  if (p >= n_parents) return;
+
+  uint4 seed;
+  if (level == 0) {
+    const int* rp = keys + key_base + 516;
+    seed = make_uint4((u32)rp[0], (u32)rp[1], (u32)rp[2], (u32)rp[3]);
+  } else {
+    seed = parents[(u64)key_id * (n_parents) + (u64)p];
+    // parents buffer is sized per level: indexed [key][p] with stride
+    // n_parents (the launcher passes the matching base pointer)
+  }
+  const int sel = (int)(seed.x & 1u);
+  uint4 c0, c1;
+  prf_pair<PRF>(seed, T, c0, c1);
+  c0 = add128(c0, cw_lvl[sel * 2 + 0]);
+  c1 = add128(c1, cw_lvl[sel * 2 + 1]);
+  if (level == depth - 1) {
+    u32* orow = out + (u64)key_id * (u64)n;
+    orow[p] = c0.x;
+    orow[p + ((u64)1 << (depth - 1))] = c1.x;
+  } else {
+    uint4* crow = children + (u64)key_id * (n_parents * 2);
+    crow[p] = c0;
+    crow[p + n_parents] = c1;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // ALU / PRF probe kernels — the unit-test analog of the reference's
 // dpf_gpu/tests/test_128_bit.cu:192-200 (device add/mul/pack asserted
 // against host __int128): one element per thread, results compared
@@ -948,6 +1020,63 @@ void launch_probe_prf(std::uintptr_t seeds, std::uintptr_t aes_tabs,
       throw std::invalid_argument("unknown PRF");
   }
   HIP_CHECK(hipGetLastError());
+}
+
+namespace {
+uint4* get_scratch(size_t bytes, hipStream_t stream);  // defined below
+
+template <int PRF>
+void launch_bfs_t(const int* keys, u32* out, const u32* aes_tabs,
+                  uint4* ping, uint4* pong, int batch, long long n, int depth,
+                  hipStream_t st) {
+  const size_t shmem =
+      (PRF == PRF_AES128 ? AES_LDS_WORDS * 4 : 0) + 4 * sizeof(uint4);
+  for (int level = 0; level < depth; ++level) {
+    const long long n_parents = (long long)1 << level;
+    const long long blocks = (n_parents + 255) / 256;
+    const uint4* parents = (level & 1) ? pong : ping;
+    uint4* children = (level & 1) ? ping : pong;
+    hipLaunchKernelGGL(dpf_bfs_level_kernel<PRF>,
+                       dim3((unsigned)blocks, (unsigned)batch), dim3(256),
+                       shmem, st, keys, parents, children, out, aes_tabs,
+                       level, depth, n);
+    HIP_CHECK(hipGetLastError());
+  }
+}
+}  // namespace
+
+void launch_bfs(std::uintptr_t keys, std::uintptr_t out,
+                std::uintptr_t aes_tabs, int batch, long long n, int depth,
+                int prf, std::uintptr_t stream) {
+  if (batch <= 0) return;
+  if (depth < 1 || ((long long)1 << depth) != n)
+    throw std::invalid_argument("bad depth/n");
+  auto st = reinterpret_cast<hipStream_t>(stream);
+  // two ping-pong frontier buffers, each batch * n/2 seeds
+  const size_t half = (size_t)batch * (size_t)(n / 2) * sizeof(uint4);
+  if (2 * half > (size_t)48 << 30)
+    throw std::invalid_argument("BFS frontier would exceed 48 GiB scratch");
+  uint4* ping = get_scratch(2 * half > 0 ? 2 * half : 16, st);
+  uint4* pong = ping + (size_t)batch * (size_t)(n / 2);
+  auto* kp = reinterpret_cast<const int*>(keys);
+  auto* op = reinterpret_cast<u32*>(out);
+  auto* ap = reinterpret_cast<const u32*>(aes_tabs);
+  switch (prf) {
+    case PRF_DUMMY:
+      launch_bfs_t<PRF_DUMMY>(kp, op, ap, ping, pong, batch, n, depth, st);
+      break;
+    case PRF_SALSA20:
+      launch_bfs_t<PRF_SALSA20>(kp, op, ap, ping, pong, batch, n, depth, st);
+      break;
+    case PRF_CHACHA20:
+      launch_bfs_t<PRF_CHACHA20>(kp, op, ap, ping, pong, batch, n, depth, st);
+      break;
+    case PRF_AES128:
+      launch_bfs_t<PRF_AES128>(kp, op, ap, ping, pong, batch, n, depth, st);
+      break;
+    default:
+      throw std::invalid_argument("unknown PRF");
+  }
 }
 
 // ---------------------------------------------------------------------------
