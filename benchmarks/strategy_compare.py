@@ -45,6 +45,8 @@ def main():
         "fused": lambda: dpf.eval_gpu(keys),
         "two_stage": lambda: dpf.eval_gpu(keys, strategy="two_stage"),
         "expand(one-hot)": lambda: dpf.eval_gpu(keys, one_hot_only=True),
+        "bfs(one-hot)": lambda: dpf.eval_gpu(keys, one_hot_only=True,
+                                             strategy="bfs"),
     }
     for name, fn in rows.items():
         ms = timed(fn)

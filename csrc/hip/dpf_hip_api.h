@@ -20,6 +20,14 @@ void launch_expand(std::uintptr_t keys, std::uintptr_t out,
                    std::uintptr_t aes_tabs, int batch, long long n, int depth,
                    int zlog, int prf, std::uintptr_t stream);
 
+// Level-synchronized breadth-first expansion (the reference's
+// dpf_breadth_first.cu strategy): depth kernel launches, ping-pong u128
+// frontiers in scratch, NATURAL-order low-32 one-hot output
+//   out: device ptr, u32 [batch][n]
+void launch_bfs(std::uintptr_t keys, std::uintptr_t out,
+                std::uintptr_t aes_tabs, int batch, long long n, int depth,
+                int prf, std::uintptr_t stream);
+
 // Naive per-leaf oracle (O(n log n) PRFs), natural order output.  Test-only.
 void launch_naive(std::uintptr_t keys, std::uintptr_t out,
                   std::uintptr_t aes_tabs, int batch, long long n, int depth,

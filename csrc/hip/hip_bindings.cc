@@ -67,6 +67,10 @@ PYBIND11_MODULE(_hip, m) {
         py::arg("out"), py::arg("aes_tabs"), py::arg("batch"), py::arg("n"),
         py::arg("depth"), py::arg("zlog"), py::arg("prf"), py::arg("stream"),
         py::call_guard<py::gil_scoped_release>());
+  m.def("eval_bfs", &gpudpf_hip::launch_bfs, py::arg("keys"), py::arg("out"),
+        py::arg("aes_tabs"), py::arg("batch"), py::arg("n"), py::arg("depth"),
+        py::arg("prf"), py::arg("stream"),
+        py::call_guard<py::gil_scoped_release>());
   m.def("eval_naive", &gpudpf_hip::launch_naive, py::arg("keys"),
         py::arg("out"), py::arg("aes_tabs"), py::arg("batch"), py::arg("n"),
         py::arg("depth"), py::arg("prf"), py::arg("stream"),

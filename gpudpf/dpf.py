@@ -277,18 +277,27 @@ class DPF(object):
             chunk = keys_gpu[lo:hi].contiguous()
             b = hi - lo
             if one_hot_only:
-                if self._perm_gpu is None:
-                    raise Exception(
-                        "one_hot_only needs the materialized row "
-                        "permutation (domain > PERM_MATERIALIZE_MAX)")
                 out = torch.empty((b, n), dtype=torch.int32, device=dev)
-                _hip.eval_expand(
-                    chunk.data_ptr(), out.data_ptr(), self._aes_ptr, b, n,
-                    depth, self._zlog, self.prf_method, stream,
-                )
-                # rows are in leaf_perm order; gather back to natural
-                # order and trim the power-of-two padding
-                out = out.index_select(1, self._perm_gpu)
+                if strategy == "bfs":
+                    # level-synchronized breadth-first expansion: writes
+                    # NATURAL-order rows directly (no un-permute gather)
+                    _hip.eval_bfs(
+                        chunk.data_ptr(), out.data_ptr(), self._aes_ptr, b,
+                        n, depth, self.prf_method, stream,
+                    )
+                else:
+                    if self._perm_gpu is None:
+                        raise Exception(
+                            "one_hot_only needs the materialized row "
+                            "permutation (domain > PERM_MATERIALIZE_MAX); "
+                            "use strategy='bfs' for natural-order output")
+                    _hip.eval_expand(
+                        chunk.data_ptr(), out.data_ptr(), self._aes_ptr, b,
+                        n, depth, self._zlog, self.prf_method, stream,
+                    )
+                    # rows are in leaf_perm order; gather back to natural
+                    # order and trim the power-of-two padding
+                    out = out.index_select(1, self._perm_gpu)
                 out = out[:, : self.table_num_entries]
             else:
                 # zeroed: the kernel's j-split segments accumulate with atomics

@@ -67,6 +67,24 @@ def test_gpu_onehot_matches_cpu():
         assert np.array_equal(got[1], want2), prf
 
 
+def test_gpu_bfs_strategy_matches_cpu():
+    # level-synchronized breadth-first expansion (natural-order output)
+    # must agree with the CPU core for every PRF, including a deep-enough
+    # tree to cross several frontier ping-pongs
+    for n in (512, 1 << 14):
+        for prf in PRFS:
+            dpf = DPF(prf=prf)
+            k1, k2 = dpf.gen(n // 3, n)
+            table = torch.zeros((n, 1), dtype=torch.int32)
+            dpf.eval_init(table)
+            got = dpf.eval_gpu([k1, k2], one_hot_only=True,
+                               strategy="bfs").numpy()
+            want1 = _core.expand(k1.numpy(), prf)
+            want2 = _core.expand(k2.numpy(), prf)
+            assert np.array_equal(got[0], want1), (n, prf)
+            assert np.array_equal(got[1], want2), (n, prf)
+
+
 def test_gpu_naive_kernel_oracle():
     from gpudpf import _hip
 
