@@ -54,6 +54,55 @@ class TransformerModel(nn.Module):
         return self.decoder(self.transformer(emb, mask)), None
 
 
+class Vocab:
+    """First-appearance word index (the reference's Dictionary,
+    language_model/data.py:8-19)."""
+
+    def __init__(self):
+        self.word2idx = {}
+        self.idx2word = []
+
+    def add(self, word):
+        if word not in self.word2idx:
+            self.word2idx[word] = len(self.idx2word)
+            self.idx2word.append(word)
+        return self.word2idx[word]
+
+    def __len__(self):
+        return len(self.idx2word)
+
+
+def tokenize_text(path, vocab=None):
+    """Whitespace-tokenize a text file, appending <eos> per line (the
+    reference's Corpus.tokenize, language_model/data.py:28-49): returns
+    (int64 token tensor, vocab).  WikiText-2's train.txt/valid.txt are in
+    exactly this format."""
+    if vocab is None:
+        vocab = Vocab()
+    ids = []
+    with open(path, "r", encoding="utf8") as f:
+        for line in f:
+            for word in line.split() + ["<eos>"]:
+                ids.append(vocab.add(word))
+    return torch.tensor(ids, dtype=torch.long), vocab
+
+
+def load_text_corpus(data_path):
+    """Real-text corpus: data_path may be a WikiText-2-style directory
+    (train.txt + valid.txt [+ test.txt], one shared vocab) or a single
+    .txt file (90/10 split).  Returns (train_tokens, val_tokens, vocab)."""
+    if os.path.isdir(data_path):
+        vocab = Vocab()
+        train, vocab = tokenize_text(os.path.join(data_path, "train.txt"),
+                                     vocab)
+        val, vocab = tokenize_text(os.path.join(data_path, "valid.txt"),
+                                   vocab)
+        return train, val, vocab
+    tokens, vocab = tokenize_text(data_path)
+    split = int(tokens.shape[0] * 0.9)
+    return tokens[:split], tokens[split:], vocab
+
+
 def _synthetic_corpus(vocab, length, seed):
     """Zipf-distributed token stream (same shape as a tokenized corpus)."""
     g = torch.Generator().manual_seed(seed)
@@ -74,14 +123,27 @@ class LanguageModelDataset:
         self.vocab = vocab
         self.bptt = bptt
         self.batch_size = batch_size
+        self.vocab_words = None
         if data_path and os.path.exists(data_path):
-            tokens = torch.load(data_path)
-            self.vocab = int(tokens.max().item()) + 1
+            if data_path.endswith(".pt"):
+                tokens = torch.load(data_path)
+                self.vocab = int(tokens.max().item()) + 1
+                split = int(tokens.shape[0] * 0.9)
+                train_tokens, val_tokens = tokens[:split], tokens[split:]
+            else:
+                # real text: WikiText-2-style directory or a .txt file
+                train_tokens, val_tokens, v = load_text_corpus(data_path)
+                self.vocab = len(v)
+                self.vocab_words = v
         else:
             tokens = _synthetic_corpus(vocab, corpus_len, seed)
-        split = int(tokens.shape[0] * 0.9)
-        self.train_data = batchify(tokens[:split], batch_size)
-        self.val_data = batchify(tokens[split:], batch_size)
+            split = int(tokens.shape[0] * 0.9)
+            train_tokens, val_tokens = tokens[:split], tokens[split:]
+        self.train_data = batchify(train_tokens, batch_size)
+        self.val_data = batchify(val_tokens, batch_size)
+        self.unk_id = min(UNK_ID, self.vocab - 1)
+        if self.vocab_words is not None and "<unk>" in self.vocab_words.word2idx:
+            self.unk_id = self.vocab_words.word2idx["<unk>"]
         cls = RNNModel if model == "lstm" else TransformerModel
         self.model = cls(self.vocab)
         self.num_entries = self.vocab
@@ -163,7 +225,7 @@ class LanguageModelDataset:
                         recovered += s["recovered"]
                         for r in range(x.shape[0]):
                             if int(x[r, col]) not in rec:
-                                x[r, col] = UNK_ID
+                                x[r, col] = self.unk_id
                 out, _ = model(x, None)
                 loss = self.criterion(out.view(-1, self.vocab), y.reshape(-1))
                 total += loss.item()

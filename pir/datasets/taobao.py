@@ -33,6 +33,84 @@ class CTRModel(nn.Module):
         return self.mlp(torch.cat(embs, dim=1)).squeeze(1)
 
 
+def _remap_columns(rows):
+    """Per-column categorical remap to dense first-appearance indices
+    (the reference's process_features_list semantics,
+    taobao_rec_dataset_v2.py:67-85)."""
+    cols = len(rows[0])
+    maps = [dict() for _ in range(cols)]
+    for r in rows:
+        for c in range(cols):
+            if r[c] not in maps[c]:
+                maps[c][r[c]] = len(maps[c])
+    remapped = [[maps[c][r[c]] for c in range(cols)] for r in rows]
+    return remapped, maps
+
+
+def load_taobao_csvs(path, sample_limit=None, hist_len=16):
+    """Real Taobao CTR pipeline (reference:
+    taobao_rec_dataset_v2.py:87-197): reads the three CSVs of the Taobao
+    display-ad dataset, remaps every categorical column to dense indices,
+    joins click events with user/ad profiles (skipping events whose
+    profile rows are missing), and accumulates each user's click-history
+    access pattern over the remapped ad ids.
+
+    ad_feature.csv:   adgroup_id,cate_id,campaign_id,customer,brand,price
+                      (brand NULL -> 0; the dense price column is dropped
+                      from the sparse feature set)
+    user_profile.csv: userid + 8 categorical features (empty -> 0)
+    raw_sample.csv:   user,time_stamp,adgroup_id,pid,nonclk,clk
+
+    Returns (samples, field_sizes, num_items): samples are
+    (fields, history_before_event, clk) with history capped at hist_len.
+    """
+    def rows_of(name, parse):
+        out = []
+        with open(os.path.join(path, name), "r", encoding="utf8") as f:
+            for i, line in enumerate(f):
+                if i == 0:
+                    continue  # header
+                parts = line.rstrip("\n").split(",")
+                out.append(parse(parts))
+        return out
+
+    def ad_parse(p):
+        brand = 0 if p[4].strip() in ("", "NULL") else int(float(p[4]))
+        return [int(p[0]), int(p[1]), int(p[2]), int(p[3]), brand]
+
+    def user_parse(p):
+        return [0 if x.strip() in ("", "NULL") else int(float(x)) for x in p]
+
+    ads_raw = rows_of("ad_feature.csv", ad_parse)
+    users_raw = rows_of("user_profile.csv", user_parse)
+    ads, ad_maps = _remap_columns(ads_raw)
+    users, user_maps = _remap_columns(users_raw)
+    ad_by_id = {r[0]: r for r in ads}
+    user_by_id = {r[0]: r for r in users}
+
+    samples = []
+    histories = {}
+    events = rows_of(
+        "raw_sample.csv",
+        lambda p: (int(p[0]), int(p[1]), int(p[2]), int(p[5])))
+    events.sort(key=lambda ev: ev[1])  # chronological
+    for uid_raw, _ts, aid_raw, clk in events:
+        if sample_limit is not None and len(samples) >= sample_limit:
+            break
+        if uid_raw not in user_maps[0] or aid_raw not in ad_maps[0]:
+            continue  # profile row missing: skip, as the reference does
+        uid = user_maps[0][uid_raw]
+        aid = ad_maps[0][aid_raw]
+        fields = ad_by_id[aid] + user_by_id[uid]
+        hist = histories.setdefault(uid, [])
+        samples.append((fields, list(hist[-hist_len:]), float(clk)))
+        hist.append(aid)  # history grows AFTER the event (no label leak)
+
+    field_sizes = [len(m) for m in ad_maps] + [len(m) for m in user_maps]
+    num_items = len(ad_maps[0])
+    return samples, field_sizes, num_items
+
+
 def _synthetic_ctr(num_items, field_sizes, num_samples, hist_len, seed):
     rng = np.random.default_rng(seed)
     w = 1.0 / np.arange(1, num_items + 1)
@@ -56,13 +134,12 @@ class TaobaoDataset:
         self.num_items = num_items
         self.field_sizes = list(field_sizes)
         if data_path and os.path.exists(data_path):
-            # a preprocessed trace would be loaded here; raw-Taobao CSV
-            # parsing is deployment-specific
-            raise NotImplementedError(
-                "supply a preprocessed trace; raw-taobao CSV parsing is "
-                "site-specific")
-        samples = _synthetic_ctr(num_items, self.field_sizes, num_samples,
-                                 hist_len, seed)
+            samples, self.field_sizes, self.num_items = load_taobao_csvs(
+                data_path, sample_limit=num_samples, hist_len=hist_len)
+            num_items = self.num_items
+        else:
+            samples = _synthetic_ctr(num_items, self.field_sizes,
+                                     num_samples, hist_len, seed)
         split = int(len(samples) * 0.8)
         self.train_samples = samples[:split]
         self.eval_samples = samples[split:]
