@@ -1,17 +1,26 @@
 """Production serving helpers.
 
 GraphedServer: a hipGraph-captured serving step for fixed-batch key
-streams — the launch-bound inner loop (zero + fused kernel) is captured
-once with torch.cuda.CUDAGraph (hipGraph on ROCm) and replayed per batch,
-leaving only the key H2D copy and result D2H on the host path.  This is
-the idiomatic MI355X replacement for the reference's per-call stream
+streams — the launch-bound inner loop is captured once with
+torch.cuda.CUDAGraph (hipGraph on ROCm) and replayed per batch, leaving
+only the key H2D copy and result D2H on the host path.  This is the
+idiomatic MI355X replacement for the reference's per-call stream
 creation (dpf_wrapper.cu:155-156).
 
-Measured at batch=512 this is throughput-NEUTRAL vs plain launches
-(0.58 vs 0.55 ms at n=16384): ROCm launch overhead for a single fused
-kernel is already small.  The value is the pinned, preallocated,
-fixed-address serving loop (graphs would pay off with many small
-launches per step).
+Serves all backend shapes:
+  * fused (entry <= 16 words): one captured zero+fused launch;
+  * wide entries: captured expand + streaming-GEMM sequence (the MFMA
+    digit-plane path allocates per call and is not graph-capturable;
+    the streaming GEMM reads the table in place, so the whole two-stage
+    pipeline replays from one graph);
+  * ShardedDPF: the local fused partial is graph-replayed and the RCCL
+    all-reduce runs outside the graph (collectives are not captured).
+
+Measured at batch=512 the fused graph is throughput-NEUTRAL vs plain
+launches (0.58 vs 0.55 ms at n=16384): ROCm launch overhead for a single
+fused kernel is already small.  The value is the pinned, preallocated,
+fixed-address serving loop, and launch-count amortization for the
+multi-launch wide path.
 """
 
 import torch
@@ -25,33 +34,68 @@ class GraphedServer:
     Usage:
         srv = GraphedServer(dpf, batch=512)   # dpf already eval_init'ed
         shares = srv.eval(keys_cpu)           # [batch, e] int32 (CPU)
+
+    `dpf` may be a DPF or a ShardedDPF (then keys passed to eval must be
+    full-domain keys; the subkey restriction runs on the host and the
+    partial-sum all-reduce runs after graph replay).
     """
 
-    def __init__(self, dpf: DPF, batch: int):
+    MAX_WIDE_SHARES_BYTES = 16 << 30
+
+    def __init__(self, dpf, batch: int):
+        self.sharded = None
+        if hasattr(dpf, "local"):  # ShardedDPF
+            self.sharded = dpf
+            dpf = dpf.local
         if dpf._table_gpu is None:
             raise Exception("eval_init the DPF before building a server")
-        if dpf._entry_padded != DPF.ENTRY_SIZE:
-            raise Exception("GraphedServer serves the fused path (e <= 16)")
         self.dpf = dpf
         self.batch = batch
         dev = dpf._table_gpu.device
         self.device = dev
+        self.wide = dpf._entry_padded > DPF.ENTRY_SIZE
+        if self.sharded is not None and self.wide:
+            raise Exception("sharded GraphedServer serves the fused path "
+                            "(entry <= 16 words)")
         self._keys_gpu = torch.zeros((batch, DPF.KEY_INTS), dtype=torch.int32,
                                      device=dev)
-        self._out_gpu = torch.zeros((batch, DPF.ENTRY_SIZE), dtype=torch.int32,
-                                    device=dev)
         self._keys_pinned = torch.zeros((batch, DPF.KEY_INTS),
                                         dtype=torch.int32).pin_memory()
-
         n = dpf._n_domain
+        ep = dpf._entry_padded
+        self._out_gpu = torch.zeros((batch, ep), dtype=torch.int32,
+                                    device=dev)
+        if self.wide:
+            shares_bytes = batch * n * 4
+            if shares_bytes > self.MAX_WIDE_SHARES_BYTES:
+                raise Exception(
+                    "wide GraphedServer share buffer would need %d bytes "
+                    "(> %d); lower the batch" %
+                    (shares_bytes, self.MAX_WIDE_SHARES_BYTES))
+            self._shares_gpu = torch.zeros((batch, n), dtype=torch.int32,
+                                           device=dev)
 
         def launch():
             stream = torch.cuda.current_stream(dev).cuda_stream
             self._out_gpu.zero_()
-            _hip.eval_fused(self._keys_gpu.data_ptr(),
-                            dpf._table_gpu.data_ptr(),
-                            self._out_gpu.data_ptr(), dpf._aes_ptr, batch, n,
-                            dpf._depth, dpf._zlog, dpf.prf_method, stream)
+            if self.wide:
+                _hip.eval_expand(self._keys_gpu.data_ptr(),
+                                 self._shares_gpu.data_ptr(), dpf._aes_ptr,
+                                 batch, n, dpf._depth, dpf._zlog,
+                                 dpf.prf_method, stream)
+                for lo in range(0, batch, 64):
+                    hi = min(batch, lo + 64)
+                    _hip.gemm_u32_stream(
+                        self._shares_gpu[lo:hi].data_ptr(),
+                        dpf._table_gpu.data_ptr(),
+                        self._out_gpu[lo:hi].data_ptr(), hi - lo, n, ep,
+                        stream)
+            else:
+                _hip.eval_fused(self._keys_gpu.data_ptr(),
+                                dpf._table_gpu.data_ptr(),
+                                self._out_gpu.data_ptr(), dpf._aes_ptr,
+                                batch, n, dpf._depth, dpf._zlog,
+                                dpf.prf_method, stream)
 
         # warm up on a side stream, then capture
         s = torch.cuda.Stream(dev)
@@ -65,11 +109,14 @@ class GraphedServer:
         with torch.cuda.graph(self._graph):
             launch()
 
-    def eval(self, keys):
-        """keys: [batch, 524] int32 CPU tensor (or list).  Returns CPU
-        [batch, e] shares."""
+    def eval(self, keys, to_host=True):
+        """keys: [batch, 524] int32 CPU tensor (or list) — full-domain
+        keys for a sharded server, which restricts them per rank.
+        Returns [batch, e] shares (CPU by default)."""
         if not isinstance(keys, torch.Tensor):
             keys = torch.stack([k.reshape(-1) for k in keys])
+        if self.sharded is not None:
+            keys = self.sharded.shard_subkeys(keys)
         if keys.shape[0] != self.batch:
             raise Exception("GraphedServer is fixed at batch=%d" % self.batch)
         # raw memcpy into the pinned staging buffer: torch's copy_ into a
@@ -78,4 +125,8 @@ class GraphedServer:
         self._keys_pinned.numpy()[:] = keys.numpy()
         self._keys_gpu.copy_(self._keys_pinned, non_blocking=True)
         self._graph.replay()
-        return self._out_gpu[:, : self.dpf.table_effective_entry_size].cpu()
+        out = self._out_gpu
+        if self.sharded is not None:
+            out = self.sharded._allreduce_(out)
+        out = out[:, : self.dpf.table_effective_entry_size]
+        return out.cpu() if to_host else out

@@ -106,6 +106,17 @@ def _nccl_worker(port, collective, q):
         rec2 = (o1.to(torch.int64) - o2.to(torch.int64)).to(
             torch.int32).cpu().numpy()
         ok = ok and bool(np.array_equal(rec2, table[idxs, :].numpy()))
+
+        # sharded GraphedServer: graph-replayed local partial + RCCL
+        # all-reduce outside the graph
+        from gpudpf.serving import GraphedServer
+
+        srv = GraphedServer(sd, batch=len(idxs))
+        sa = srv.eval(torch.stack([k[0] for k in ks]))
+        sb = srv.eval(torch.stack([k[1] for k in ks]))
+        rec3 = (sa.to(torch.int64) - sb.to(torch.int64)).to(
+            torch.int32).numpy()
+        ok = ok and bool(np.array_equal(rec3, table[idxs, :].numpy()))
         q.put((0, ok, ""))
     except Exception as e:  # pragma: no cover
         q.put((0, False, repr(e)))

@@ -262,6 +262,56 @@ def test_gpu_graphed_server_matches_eval():
     assert torch.equal(srv.eval(kt2), dpf.eval_gpu(kt2))
 
 
+def test_gpu_graphed_server_wide_entries():
+    # wide path: captured expand + streaming-GEMM graph, incl. >64-batch
+    # (two GEMM launches in one graph)
+    from gpudpf.serving import GraphedServer
+
+    n, e, batch = 8192, 40, 96
+    dpf = DPF(prf=DPF.PRF_CHACHA20)
+    table = torch.randint(-(2**31), 2**31 - 1, (n, e), dtype=torch.int64).to(
+        torch.int32)
+    dpf.eval_init(table)
+    srv = GraphedServer(dpf, batch)
+    assert srv.wide
+    keys = torch.stack([dpf.gen((i * 53) % n, n)[0] for i in range(batch)])
+    want = dpf.eval_gpu(keys)
+    got = srv.eval(keys)
+    assert torch.equal(got, want)
+    keys2 = torch.stack([dpf.gen((i * 7 + 1) % n, n)[0] for i in range(batch)])
+    assert torch.equal(srv.eval(keys2), dpf.eval_gpu(keys2))
+
+
+def test_gpu_graphed_server_scratch_regrow_safe():
+    # Capture a graph at a small domain, then run a much deeper eval that
+    # forces the shared scratch buffer to grow: the captured graph must
+    # still replay correctly (round-1 advisor finding — the old code freed
+    # the captured allocation).
+    from gpudpf.serving import GraphedServer
+
+    n_small, batch = 16384, 32
+    dpf = DPF(prf=DPF.PRF_SALSA20)
+    table = torch.randint(-(2**31), 2**31 - 1, (n_small, 16),
+                          dtype=torch.int64).to(torch.int32)
+    dpf.eval_init(table)
+    srv = GraphedServer(dpf, batch)
+    keys = torch.stack([dpf.gen(i % n_small, n_small)[0]
+                        for i in range(batch)])
+    want = srv.eval(keys)
+
+    # deep eval on a second instance grows the per-device scratch
+    n_big = 1 << 22
+    dpf2 = DPF(prf=DPF.PRF_SALSA20)
+    t2 = torch.zeros((n_big, 16), dtype=torch.int32)
+    dpf2.eval_init(t2)
+    k1, _ = dpf2.gen(5, n_big)
+    dpf2.eval_gpu([k1])
+
+    # the original graph still replays against live memory
+    got = srv.eval(keys)
+    assert torch.equal(got, want)
+
+
 def test_gpu_deep_tree_correctness():
     # n=2^22: DS=14 -> 7 global-scratch stack levels + 4 LDS + 2 register
     # levels all exercised; reconstruction vs ground truth

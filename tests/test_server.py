@@ -46,3 +46,35 @@ def test_pir_http_roundtrip():
     b = post(k2s)
     rec = (a.to(torch.int64) - b.to(torch.int64)).to(torch.int32)
     assert torch.equal(rec, table[idxs, :])
+
+
+@pytest.mark.gpu
+def test_pir_http_roundtrip_gpu():
+    """Same HTTP layer exercising the GPU eval path (build_app routes to
+    eval_gpu when a GPU is present)."""
+    assert torch.cuda.is_available()
+    n, e = 4096, 16
+    torch.manual_seed(13)
+    table = torch.randint(-(2**31), 2**31 - 1, (n, e), dtype=torch.int64).to(
+        torch.int32)
+    app = build_app(table=table, prf=DPF.PRF_AES128)
+    client = TestClient(app)
+    assert client.get("/info").json()["device"].startswith("cuda")
+
+    dpf = DPF(prf=DPF.PRF_AES128)
+    idxs = [0, 4095, 1234]
+    pairs = [dpf.gen(i, n) for i in idxs]
+
+    def post(keys):
+        blob = base64.b64encode(
+            torch.stack(keys).numpy().astype(np.int32).tobytes()).decode()
+        r = client.post("/eval", json={"keys_b64": blob})
+        assert r.status_code == 200
+        raw = base64.b64decode(r.json()["shares_b64"])
+        return torch.from_numpy(
+            np.frombuffer(raw, dtype=np.int32).reshape(len(keys), -1).copy())
+
+    a = post([p[0] for p in pairs])
+    b = post([p[1] for p in pairs])
+    rec = (a.to(torch.int64) - b.to(torch.int64)).to(torch.int32)
+    assert torch.equal(rec, table[idxs, :])
