@@ -355,8 +355,64 @@ __device__ __forceinline__ void aes_cipher_pair(uint4 seed, const AesLds& T,
   u32 s0 = k0, s1 = k1, s2 = k2, s3 = k3;
   u32 u0 = (1u << 24) ^ k0, u1 = k1, u2 = k2, u3 = k3;
   u32 rcon = 0x01u;
+  int r_begin = 1;
+#ifndef GPUDPF_AES_NOSHARE
+  // Shared-prefix rounds: the two children's plaintexts (pos 0 / pos 1)
+  // differ in a single byte, so their states agree except for byte
+  // (s0 >> 24) until MixColumns diffuses it — round 1 of child b costs
+  // ONE extra T-lookup (all four output words share), and round 2 shares
+  // 12 of its 16 lookups (each output word reads exactly one byte of the
+  // diverged word 0).  Saves ~27 of ~320 lookups per pair; rounds 3+ are
+  // fully diverged and run the common loop below.  A/B toggle:
+  // compile with -DGPUDPF_AES_NOSHARE for the straight dual-cipher
+  // variant (measured comparison in profiles/PROFILING.md).
+  {
+    // round 1
+    u32 w_ = (k3 << 8) | (k3 >> 24);
+    w_ = (T.sbox((w_ >> 24) & 0xff) << 24) | (T.sbox((w_ >> 16) & 0xff) << 16) |
+         (T.sbox((w_ >> 8) & 0xff) << 8) | T.sbox(w_ & 0xff);
+    w_ ^= (rcon << 24);
+    rcon = 0x02u;
+    k0 ^= w_; k1 ^= k0; k2 ^= k1; k3 ^= k2;
+    const u32 A = T.te0(s0 >> 24);
+    u32 n0 = A ^ rotr8(T.te0((s1 >> 16) & 0xff)) ^
+             rotr16(T.te0((s2 >> 8) & 0xff)) ^ rotr24(T.te0(s3 & 0xff)) ^ k0;
+    u32 n1 = T.te0(s1 >> 24) ^ rotr8(T.te0((s2 >> 16) & 0xff)) ^
+             rotr16(T.te0((s3 >> 8) & 0xff)) ^ rotr24(T.te0(s0 & 0xff)) ^ k1;
+    u32 n2 = T.te0(s2 >> 24) ^ rotr8(T.te0((s3 >> 16) & 0xff)) ^
+             rotr16(T.te0((s0 >> 8) & 0xff)) ^ rotr24(T.te0(s1 & 0xff)) ^ k2;
+    u32 n3 = T.te0(s3 >> 24) ^ rotr8(T.te0((s0 >> 16) & 0xff)) ^
+             rotr16(T.te0((s1 >> 8) & 0xff)) ^ rotr24(T.te0(s2 & 0xff)) ^ k3;
+    const u32 m0 = n0 ^ A ^ T.te0((s0 >> 24) ^ 1u);
+    // round 2
+    w_ = (k3 << 8) | (k3 >> 24);
+    w_ = (T.sbox((w_ >> 24) & 0xff) << 24) | (T.sbox((w_ >> 16) & 0xff) << 16) |
+         (T.sbox((w_ >> 8) & 0xff) << 8) | T.sbox(w_ & 0xff);
+    w_ ^= (rcon << 24);
+    rcon = 0x04u;
+    k0 ^= w_; k1 ^= k0; k2 ^= k1; k3 ^= k2;
+    const u32 c0 = rotr8(T.te0((n1 >> 16) & 0xff)) ^
+                   rotr16(T.te0((n2 >> 8) & 0xff)) ^
+                   rotr24(T.te0(n3 & 0xff)) ^ k0;
+    const u32 c1 = T.te0(n1 >> 24) ^ rotr8(T.te0((n2 >> 16) & 0xff)) ^
+                   rotr16(T.te0((n3 >> 8) & 0xff)) ^ k1;
+    const u32 c2 = T.te0(n2 >> 24) ^ rotr8(T.te0((n3 >> 16) & 0xff)) ^
+                   rotr24(T.te0(n1 & 0xff)) ^ k2;
+    const u32 c3 = T.te0(n3 >> 24) ^ rotr16(T.te0((n1 >> 8) & 0xff)) ^
+                   rotr24(T.te0(n2 & 0xff)) ^ k3;
+    s0 = T.te0(n0 >> 24) ^ c0;
+    u0 = T.te0(m0 >> 24) ^ c0;
+    s1 = rotr24(T.te0(n0 & 0xff)) ^ c1;
+    u1 = rotr24(T.te0(m0 & 0xff)) ^ c1;
+    s2 = rotr16(T.te0((n0 >> 8) & 0xff)) ^ c2;
+    u2 = rotr16(T.te0((m0 >> 8) & 0xff)) ^ c2;
+    s3 = rotr8(T.te0((n0 >> 16) & 0xff)) ^ c3;
+    u3 = rotr8(T.te0((m0 >> 16) & 0xff)) ^ c3;
+    r_begin = 3;
+  }
+#endif
 #pragma unroll
-  for (int r = 1; r < 10; ++r) {
+  for (int r = r_begin; r < 10; ++r) {
     // next round key (k0..k3 become rk[4r..4r+3])
     u32 w_ = (k3 << 8) | (k3 >> 24);
     w_ = (T.sbox((w_ >> 24) & 0xff) << 24) | (T.sbox((w_ >> 16) & 0xff) << 16) |
