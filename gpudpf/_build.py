@@ -84,11 +84,13 @@ def build_hip(force=False):
     if not force and not _needs_build(target, sources):
         return target
     hipcc = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
+    extra = os.environ.get("GPUDPF_HIP_EXTRA_FLAGS", "").split()
     cmd = (
         [hipcc, "-O3", "-std=c++17", "-shared", "-fPIC",
          "--offload-arch=" + GFX_ARCH, "-fvisibility=hidden",
          "-I" + os.path.join(CSRC, "core"),
          "-x", "hip"]
+        + extra
         + ["-I" + i for i in _pybind_includes()]
         + sources
         + ["-o", target]
