@@ -27,14 +27,18 @@ QUICK_GRID = {
 }
 
 
-def _make_dataset(name, size):
-    """size: quick (smoke), medium (full grid in minutes), full."""
+def _make_dataset(name, size, data_path=None):
+    """size: quick (smoke), medium (full grid in minutes), full.
+    data_path: real dataset location (WikiText-2-style dir / Taobao CSV
+    dir / ml-20m ratings.csv); synthetic proxies when absent."""
     if name == "lm":
         from pir.datasets import language_model
 
         kw = {"quick": dict(vocab=512, corpus_len=20000),
               "medium": dict(vocab=1024, corpus_len=60000),
               "full": {}}[size]
+        if data_path:
+            kw = dict(data_path=data_path)
         ds = language_model.initialize(**kw)
         mb = {"quick": 20, "medium": 80, "full": None}[size]
         ds.train_model(epochs=1, max_batches=mb)
@@ -44,6 +48,8 @@ def _make_dataset(name, size):
         kw = {"quick": dict(num_items=512, num_users=400),
               "medium": dict(num_items=2048, num_users=1500),
               "full": {}}[size]
+        if data_path:
+            kw = dict(data_path=data_path)
         ds = movielens.initialize(**kw)
         ds.train_model(epochs=1 if size == "quick" else 2)
     elif name == "taobao":
@@ -52,6 +58,10 @@ def _make_dataset(name, size):
         kw = {"quick": dict(num_items=512, num_samples=500),
               "medium": dict(num_items=4096, num_samples=2000),
               "full": {}}[size]
+        if data_path:
+            kw = dict(data_path=data_path,
+                      num_samples={"quick": 2000, "medium": 20000,
+                                   "full": None}[size])
         ds = taobao.initialize(**kw)
         ds.train_model(epochs=1 if size == "quick" else 2)
     else:
@@ -62,12 +72,12 @@ def _make_dataset(name, size):
 _DS = None
 
 
-def _init_worker(name, size):
+def _init_worker(name, size, data_path=None):
     global _DS
     import torch
 
     torch.manual_seed(0)
-    _DS = _make_dataset(name, size)
+    _DS = _make_dataset(name, size, data_path)
 
 
 def run_config(args):
@@ -103,6 +113,9 @@ def main():
     ap.add_argument("--size", default=None,
                     choices=["quick", "medium", "full"])
     ap.add_argument("--processes", type=int, default=8)
+    ap.add_argument("--data-path", default=None,
+                    help="real dataset location (lm: WikiText-2 dir or "
+                         ".txt; taobao: CSV dir; movielens: ratings.csv)")
     a = ap.parse_args()
     size = a.size or ("quick" if a.quick else "full")
     grid = QUICK_GRID if size == "quick" else GRID
@@ -112,11 +125,11 @@ def main():
     cfgs = [dict(zip(keys, vals)) for vals in
             itertools.product(*(grid[k] for k in keys))]
     if a.processes <= 1:
-        _init_worker(a.dataset, size)
+        _init_worker(a.dataset, size, a.data_path)
         results = [run_config((c, out_dir)) for c in cfgs]
     else:
         with Pool(a.processes, initializer=_init_worker,
-                  initargs=(a.dataset, size)) as pool:
+                  initargs=(a.dataset, size, a.data_path)) as pool:
             results = pool.map(run_config, [(c, out_dir) for c in cfgs])
     print("wrote %d configs to %s" % (len(results), out_dir))
 

@@ -114,3 +114,18 @@ def test_movielens_real_csv():
     ds.train_model(epochs=1)
     res = ds.evaluate()
     assert res["metric"] == "auc"
+
+
+def test_sweep_single_config_on_real_corpus(tmp_path):
+    # the full sweep pipeline (dataset -> optimizer -> recovery ->
+    # accuracy) over the real-text corpus path
+    from pir import sweep
+
+    sweep._init_worker("lm", "quick", os.path.join(FIX, "wikitext"))
+    cfg = {"hot_fraction": 0.1, "group_size": 2, "num_bins": 8,
+           "queries_per_bin": 1}
+    res = sweep.run_config((cfg, str(tmp_path)))
+    assert "accuracy" in res and res["accuracy"]["metric"] == "ppl"
+    assert 0.0 <= res["recovery"]["recovery_rate"] <= 1.0
+    assert os.path.exists(os.path.join(
+        str(tmp_path), "hf0.1_g2_b8_q1.json"))
