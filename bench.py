@@ -73,6 +73,9 @@ def main():
     ap.add_argument("--check", action="store_true",
                     help="verify reconstruction correctness after timing "
                          "(adds a second key batch eval)")
+    ap.add_argument("--no-graph", action="store_true",
+                    help="plain per-step launches instead of the hipGraph "
+                         "serving loop")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -115,8 +118,14 @@ def main():
         engine = ShardedDPF(prf=prf, device=device)
         engine.eval_init(table)
         if on_gpu:
-            def step(k=keys_cpu):
-                return engine.eval_gpu(k)
+            if args.no_graph:
+                def step(k=keys_cpu):
+                    return engine.eval_gpu(k)
+            else:
+                from gpudpf.serving import GraphedServer
+                srv = GraphedServer(engine, args.batch)
+                def step(k=keys_cpu):
+                    return srv.eval(k)
         else:
             def step(k=keys_cpu):
                 return engine.eval_cpu(k)
@@ -124,8 +133,16 @@ def main():
         engine = DPF(prf=prf, device=device)
         engine.eval_init(table)
         if on_gpu:
-            def step(k=keys_cpu):
-                return engine.eval_gpu(k)
+            if args.no_graph:
+                def step(k=keys_cpu):
+                    return engine.eval_gpu(k)
+            else:
+                # production serving loop: hipGraph replay of the fused
+                # step (same work per step: key H2D, kernel, share D2H)
+                from gpudpf.serving import GraphedServer
+                srv = GraphedServer(engine, args.batch)
+                def step(k=keys_cpu):
+                    return srv.eval(k)
         else:
             def step(k=keys_cpu):
                 return engine.eval_cpu(k)

@@ -82,19 +82,16 @@ class ShardedDPF(object):
     def shard_subkeys(self, keys):
         """Restrict a batch of full-domain wire-format keys to this rank's
         residue class: [b, 524] int32 CPU tensor of depth-log2(W) subkeys.
-        Host-side (the restriction walks log2(W) PRF levels per key); on a
-        serving path do this once per batch, then feed eval_gpu_into."""
-        if isinstance(keys, torch.Tensor) and keys.dim() == 2:
-            keys = [keys[i] for i in range(keys.shape[0])]
-        subs = [
-            torch.from_numpy(
-                _core.shard_subkey(
-                    k.reshape(-1).numpy(), self.prf_method, self.rank, self.world
-                )
-            )
-            for k in keys
-        ]
-        return torch.stack(subs)
+        Host-side (the restriction walks log2(W) PRF levels per key), one
+        C++ call for the whole batch; on a serving path do this once per
+        batch, then feed eval_gpu_into."""
+        if not isinstance(keys, torch.Tensor):
+            keys = torch.stack([k.reshape(-1) for k in keys])
+        if keys.dim() == 1:
+            keys = keys.unsqueeze(0)
+        return torch.from_numpy(
+            _core.shard_subkey_batch(keys.contiguous().numpy(),
+                                     self.prf_method, self.rank, self.world))
 
     # backwards-compatible internal alias
     _subkeys = shard_subkeys

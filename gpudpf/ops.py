@@ -58,10 +58,13 @@ def pir_matmul_u32_stream(shares, table, out=None):
 
     shares: [M, K] int32, table: [K, N] int32 (both already on the GPU for
     the huge-table path).  Unlike pir_matmul_u32 this allocates NO copy of
-    the table (no transpose, no digit planes) — the only extra memory is
-    the [M, N] output — so it serves tables sized to HBM capacity
-    (200+ GB).  Batches > 64 are evaluated in 64-row chunks, each chunk
-    re-streaming the table once."""
+    the table (no transpose, no digit planes) — the extra memory is the
+    [M, N] output plus a <=1 GiB K-split partials scratch — so it serves
+    tables sized to HBM capacity (200+ GB).  Batches > 16 are evaluated
+    in 16-row chunks, each chunk re-streaming the table once: 16 is the
+    bandwidth-bound register limit (4 uint4 accumulator sets per lane);
+    re-streaming at the HBM line beats one instruction-bound pass, and
+    genuinely compute-bound shapes belong to pir_matmul_u32 (MFMA)."""
     assert shares.dtype == torch.int32 and table.dtype == torch.int32
     M, K = shares.shape
     K2, N = table.shape
@@ -72,8 +75,8 @@ def pir_matmul_u32_stream(shares, table, out=None):
     c = out if out is not None else torch.empty(
         (M, N), dtype=torch.int32, device=dev)
     stream = torch.cuda.current_stream(dev).cuda_stream
-    for lo in range(0, M, 64):
-        hi = min(M, lo + 64)
+    for lo in range(0, M, 16):
+        hi = min(M, lo + 16)
         chunk = c[lo:hi]
         chunk.zero_()
         _hip.gemm_u32_stream(a[lo:hi].data_ptr(), b.data_ptr(),

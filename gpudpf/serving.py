@@ -83,8 +83,8 @@ class GraphedServer:
                                  self._shares_gpu.data_ptr(), dpf._aes_ptr,
                                  batch, n, dpf._depth, dpf._zlog,
                                  dpf.prf_method, stream)
-                for lo in range(0, batch, 64):
-                    hi = min(batch, lo + 64)
+                for lo in range(0, batch, 16):
+                    hi = min(batch, lo + 16)
                     _hip.gemm_u32_stream(
                         self._shares_gpu[lo:hi].data_ptr(),
                         dpf._table_gpu.data_ptr(),

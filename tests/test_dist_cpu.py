@@ -60,3 +60,22 @@ def test_sharded_eval_matches_single(world):
         p.join(timeout=60)
     for rank, ok, err in results:
         assert ok, f"rank {rank}: {err}"
+
+
+def test_shard_subkey_batch_matches_single():
+    # one C++ call for the whole batch == per-key restriction
+    from gpudpf import DPF, _core
+    import numpy as np
+
+    n, world = 1 << 14, 4
+    prf = DPF.PRF_CHACHA20
+    keys = []
+    for i in range(8):
+        k1, _ = _core.gen((i * 977) % n, n, b"sb-%d" % i, prf)
+        keys.append(k1)
+    kt = np.stack(keys)
+    for rank in range(world):
+        got = _core.shard_subkey_batch(kt, prf, rank, world)
+        for i in range(8):
+            want = _core.shard_subkey(keys[i], prf, rank, world)
+            assert np.array_equal(got[i], want), (rank, i)
