@@ -191,6 +191,51 @@ py::array_t<std::int64_t> leaf_perm_rows(
   return out;
 }
 
+// Compact wire format: the 524-int format always carries 64 cw slots; a
+// depth-d key only uses 2d of them, so shallow-tree keys ship dead
+// bytes (the reference's SizeOf, dpf_base/dpf.h:474, counts the same
+// padding).  Layout: [depth u128][cw0[0..2d) u128][cw1[0..2d) u128]
+// [root u128][n u128] = (3 + 4d) * 16 bytes; depth 14 -> 944 B vs 2096.
+// Lossless round-trip with the standard format; same scheme, same keys.
+py::array_t<std::int32_t> key_compact(const KeyArr& key) {
+  DpfKey k = key_from_array(key);
+  const int d = k.depth;
+  py::array_t<std::int32_t> out((py::ssize_t)((3 + 4 * d) * 4));
+  std::int32_t* op = out.mutable_data();
+  u128* slots = reinterpret_cast<u128*>(op);
+  slots[0] = (u128)(unsigned)d;
+  for (int i = 0; i < 2 * d; ++i) {
+    slots[1 + i] = k.cw[0][i];
+    slots[1 + 2 * d + i] = k.cw[1][i];
+  }
+  slots[1 + 4 * d] = k.root;
+  slots[2 + 4 * d] = (u128)k.n;
+  return out;
+}
+
+KeyArr key_expand_compact(
+    py::array_t<std::int32_t, py::array::c_style | py::array::forcecast> c) {
+  if (c.size() < 8 || c.size() % 4 != 0)
+    throw std::invalid_argument("corrupt compact key: bad length");
+  const u128* slots = reinterpret_cast<const u128*>(c.data());
+  const int d = (int)(u64)slots[0];
+  if (d < 1 || d > kMaxDepth)
+    throw std::invalid_argument("corrupt compact key: bad depth");
+  if (c.size() != (py::ssize_t)((3 + 4 * d) * 4))
+    throw std::invalid_argument("corrupt compact key: length != 16*(3+4d)");
+  DpfKey k;
+  k.depth = d;
+  for (int i = 0; i < 2 * d; ++i) {
+    k.cw[0][i] = slots[1 + i];
+    k.cw[1][i] = slots[1 + 2 * d + i];
+  }
+  k.root = slots[1 + 4 * d];
+  k.n = (u64)slots[2 + 4 * d];
+  if (k.n != ((u64)1 << d))
+    throw std::invalid_argument("corrupt compact key: n != 2^depth");
+  return key_to_array(k);
+}
+
 py::tuple prf(int method, u64 seed_lo, u64 seed_hi, u64 pos) {
   u128 s = ((u128)seed_hi << 64) | seed_lo;
   u128 r = prf_eval(method, s, (u128)pos);
@@ -341,6 +386,8 @@ PYBIND11_MODULE(_core, m) {
   m.def("aes_block", &aes_block);
   m.def("shard_subkey", &shard_subkey);
   m.def("shard_subkey_batch", &shard_subkey_batch);
+  m.def("key_compact", &key_compact);
+  m.def("key_expand_compact", &key_expand_compact);
   m.def("aes_gpu_tables", &aes_gpu_tables);
   m.def("gemm128_cpu", &gemm128_cpu);
   m.def("grid_gen", &grid_gen, py::arg("alpha"), py::arg("n_keys"),

@@ -93,6 +93,19 @@ class DPF(object):
             self.prf_method)
         return torch.from_numpy(k1s), torch.from_numpy(k2s)
 
+    @staticmethod
+    def key_compact(key):
+        """Compact wire form of a key: drops the fixed format's unused
+        correction-word slots — (3 + 4*depth) * 16 bytes, e.g. 496 B at
+        depth 7 vs the fixed 2096.  Lossless; `key_expand` restores the
+        standard 524-int form the evaluators consume."""
+        return torch.from_numpy(_core.key_compact(key.reshape(-1).numpy()))
+
+    @staticmethod
+    def key_expand(compact):
+        return torch.from_numpy(
+            _core.key_expand_compact(compact.reshape(-1).numpy()))
+
     # ------------------------------------------------------------------
     # Server side
     # ------------------------------------------------------------------

@@ -153,3 +153,19 @@ def test_large_domain_header_decode():
     d = DPF(prf=DPF.PRF_DUMMY, device="cpu")
     kt, n, depth = d._keys_tensor(torch.from_numpy(k1).unsqueeze(0))
     assert n == 1 << 32 and depth == 32
+
+
+def test_compact_key_round_trip():
+    d = DPF(prf=DPF.PRF_CHACHA20, device="cpu")
+    for n, depth in ((128, 7), (1 << 14, 14)):
+        k1, k2 = d.gen(n // 2, n)
+        c = DPF.key_compact(k1)
+        assert c.numel() * 4 == (3 + 4 * depth) * 16
+        back = DPF.key_expand(c)
+        assert torch.equal(back, k1)
+        # expanded key evaluates identically
+        a = d.eval_cpu([back], one_hot_only=True)
+        b = d.eval_cpu([k1], one_hot_only=True)
+        assert torch.equal(a, b)
+    with pytest.raises(Exception):
+        DPF.key_expand(torch.zeros(16, dtype=torch.int32))
