@@ -133,3 +133,46 @@ def test_sharded_rccl_branch_executes(collective):
     rank, ok, err = q.get(timeout=300)
     p.join(timeout=60)
     assert ok, f"rank {rank}: {err}"
+
+
+def _nccl_wide_worker(port, q):
+    """Sharded + wide entries (two-stage streaming-GEMM local path) under
+    a real nccl process group."""
+    import torch.distributed as td
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    td.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        from gpudpf import DPF, ShardedDPF, _core
+
+        N, e = 1 << 13, 40
+        prf = DPF.PRF_SALSA20
+        torch.manual_seed(5)
+        table = torch.randint(-(2**31), 2**31 - 1, (N, e),
+                              dtype=torch.int64).to(torch.int32)
+        idxs = [1, 4095, 8000]
+        sd = ShardedDPF(prf=prf, device="cuda:0")
+        sd.eval_init(table)
+        ks = [(_core.gen(i, N, b"w-%d" % i, prf)) for i in idxs]
+        a = sd.eval_gpu([torch.from_numpy(k[0]) for k in ks], to_host=False)
+        assert a.is_cuda and a.shape == (3, e)
+        b = sd.eval_gpu([torch.from_numpy(k[1]) for k in ks], to_host=False)
+        rec = (a.to(torch.int64) - b.to(torch.int64)).to(
+            torch.int32).cpu().numpy()
+        ok = bool(np.array_equal(rec, table[idxs, :].numpy()))
+        q.put((0, ok, ""))
+    except Exception as e:  # pragma: no cover
+        q.put((0, False, repr(e)))
+    finally:
+        td.destroy_process_group()
+
+
+def test_sharded_wide_entries_rccl():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    p = ctx.Process(target=_nccl_wide_worker, args=(29799, q))
+    p.start()
+    rank, ok, err = q.get(timeout=300)
+    p.join(timeout=60)
+    assert ok, f"rank {rank}: {err}"
