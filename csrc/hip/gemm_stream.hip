@@ -93,6 +93,12 @@ __global__ __launch_bounds__(kThreads) void gemm_u32_stream_kernel(
     const u32* __restrict__ a, const u32* __restrict__ b,
     u32* __restrict__ part, int batch, long long K, long long N,
     long long k_seg, long long col_tiles) {
+  // One wave64 per workgroup: LDS writes and reads of the same wave are
+  // in program order, so the A-chunk staging needs NO barriers, and the
+  // NEXT chunk's staging loads issue (into registers) before the current
+  // chunk's MAC loop — global A latency hides under the table stream.
+  constexpr int KQ = kChunkK / 4;                       // uint4s per row
+  constexpr int STG = (B * KQ + kThreads - 1) / kThreads;  // loads/lane
   __shared__ u32 s_a[B][kChunkK];
   const int t = (int)threadIdx.x;
   const long long col = (long long)blockIdx.y * kColsPerTile + 4 * t;
@@ -103,31 +109,78 @@ __global__ __launch_bounds__(kThreads) void gemm_u32_stream_kernel(
 #pragma unroll
   for (int j = 0; j < B; ++j) ax[j] = ay[j] = az[j] = aw[j] = 0;
 
-  // fast full-uint4 path needs 16-byte row alignment (N % 4 == 0; ep is
+  // fast paths need 16-byte alignment (multiple-of-4 row widths; ep is
   // always a multiple of 16 in production)
   const bool in_n = ((N & 3) == 0) && (col + 3 < N);
+  const bool a_aligned = (K & 3) == 0;
+
+  auto load_chunk = [&](long long kc, uint4* regs) {
+#pragma unroll
+    for (int s = 0; s < STG; ++s) {
+      const int i = t + s * kThreads;
+      uint4 v = make_uint4(0u, 0u, 0u, 0u);
+      if (i < B * KQ) {
+        const int j = i / KQ, kq = i % KQ;
+        const long long kk0 = kc + 4 * kq;
+        if (j < batch && kk0 < k1) {
+          const u32* arow = a + (u64)j * (u64)K + (u64)kk0;
+          if (a_aligned && kk0 + 3 < k1) {
+            v = *reinterpret_cast<const uint4*>(arow);
+          } else {
+            v.x = arow[0];
+            if (kk0 + 1 < k1) v.y = arow[1];
+            if (kk0 + 2 < k1) v.z = arow[2];
+            if (kk0 + 3 < k1) v.w = arow[3];
+          }
+        }
+      }
+      regs[s] = v;
+    }
+  };
+  auto commit_chunk = [&](const uint4* regs) {
+#pragma unroll
+    for (int s = 0; s < STG; ++s) {
+      const int i = t + s * kThreads;
+      if (i < B * KQ)
+        reinterpret_cast<uint4*>(&s_a[0][0])[i] = regs[s];
+    }
+  };
+
+  uint4 regs[STG];
+  load_chunk(k0, regs);
+  commit_chunk(regs);
+
   for (long long kc = k0; kc < k1; kc += kChunkK) {
     const int kn = (int)((kc + kChunkK <= k1) ? kChunkK : (k1 - kc));
-    __syncthreads();
-    // cooperative stage of A[:, kc:kc+kn] (rows beyond `batch` are 0)
-    for (int i = t; i < B * kChunkK; i += kThreads) {
-      const int j = i / kChunkK, kk = i % kChunkK;
-      s_a[j][kk] = (j < batch && kk < kn)
-                       ? a[(u64)j * (u64)K + (u64)(kc + kk)]
-                       : 0u;
-    }
-    __syncthreads();
+    // issue next chunk's A loads now; they retire during the MAC loop
+    const bool have_next = kc + kChunkK < k1;
+    if (have_next) load_chunk(kc + kChunkK, regs);
     if (in_n) {
       const u32* brow = b + (u64)kc * (u64)N + (u64)col;
-      for (int kk = 0; kk < kn; ++kk) {
-        const uint4 v = *reinterpret_cast<const uint4*>(brow + (u64)kk * N);
+      if (kn == kChunkK) {
+#pragma unroll 4
+        for (int kk = 0; kk < kChunkK; ++kk) {
+          const uint4 v = *reinterpret_cast<const uint4*>(brow + (u64)kk * N);
 #pragma unroll
-        for (int j = 0; j < B; ++j) {
-          const u32 s = s_a[j][kk];
-          ax[j] += s * v.x;
-          ay[j] += s * v.y;
-          az[j] += s * v.z;
-          aw[j] += s * v.w;
+          for (int j = 0; j < B; ++j) {
+            const u32 s = s_a[j][kk];
+            ax[j] += s * v.x;
+            ay[j] += s * v.y;
+            az[j] += s * v.z;
+            aw[j] += s * v.w;
+          }
+        }
+      } else {
+        for (int kk = 0; kk < kn; ++kk) {
+          const uint4 v = *reinterpret_cast<const uint4*>(brow + (u64)kk * N);
+#pragma unroll
+          for (int j = 0; j < B; ++j) {
+            const u32 s = s_a[j][kk];
+            ax[j] += s * v.x;
+            ay[j] += s * v.y;
+            az[j] += s * v.z;
+            aw[j] += s * v.w;
+          }
         }
       }
     } else if (col < N) {  // ragged tail columns, scalar loads
@@ -149,6 +202,7 @@ __global__ __launch_bounds__(kThreads) void gemm_u32_stream_kernel(
         }
       }
     }
+    if (have_next) commit_chunk(regs);
   }
   // per-segment partials: part[seg][j][N-tilewise] — plain stores
   u32* prow = part + ((u64)blockIdx.x * (u64)batch) * (u64)N;
