@@ -51,7 +51,8 @@ class CollocateConfig(NamedTuple):
 class PIRConfig(NamedTuple):
     num_bins: int = 16
     queries_per_bin: int = 1
-    key_size_model: str = "gpudpf"  # "gpudpf" (2096 B const) or "logn"
+    key_size_model: str = "gpudpf"  # "gpudpf" (2096 B const), "compact"
+                                    # (16*(3+4*depth) B), or "logn"
 
 
 KEY_BYTES = 2096
@@ -230,9 +231,14 @@ class BatchPIROptimize:
 
     # -- cost model --------------------------------------------------------
     def key_bytes(self, bin_entries: int) -> int:
+        import math
+
         if self.pir.key_size_model == "gpudpf":
             return KEY_BYTES
-        import math
+        depth = max(7, math.ceil(math.log2(max(128, bin_entries))))
+        if self.pir.key_size_model == "compact":
+            # gpudpf compact wire form (DPF.key_compact): 16*(3+4*depth) B
+            return 16 * (3 + 4 * depth)
         return 16 * 4 * max(1, math.ceil(math.log2(max(2, bin_entries))))
 
     def communication_bytes(self, entry_bytes: int = 64) -> int:
