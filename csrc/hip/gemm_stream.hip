@@ -69,6 +69,10 @@ constexpr int kMaxB = 16;          // accumulators = 4*B VGPRs per lane
 
 // Grow-only per-device partials scratch (kept alive forever so hipGraphs
 // capturing a launch stay valid — same policy as dpf_kernels.hip).
+// Shared across launches of one device: concurrent stream-GEMM launches
+// on DIFFERENT streams would race on it — the python API serializes
+// launches per call, which is the supported pattern (as with the DFS
+// scratch in dpf_kernels.hip).
 std::mutex g_part_mu;
 void* g_part[64] = {};
 size_t g_part_bytes[64] = {};
