@@ -48,11 +48,17 @@ def main():
         "bfs(one-hot)": lambda: dpf.eval_gpu(keys, one_hot_only=True,
                                              strategy="bfs"),
     }
+    # single-key latency strategies (batch=1): j-split fused vs the
+    # grid-synchronized cooperative kernel
+    one = keys[:1]
+    rows["fused(batch=1)"] = lambda: dpf.eval_gpu(one)
+    rows["coop(batch=1)"] = lambda: dpf.eval_gpu(one, strategy="coop")
     for name, fn in rows.items():
         ms = timed(fn)
-        print({"strategy": name, "prf": a.prf, "n": a.n, "batch": a.batch,
+        nb = 1 if "batch=1" in name else a.batch
+        print({"strategy": name, "prf": a.prf, "n": a.n, "batch": nb,
                "ms_per_batch": round(ms, 3),
-               "dpfs_per_sec": round(a.batch / ms * 1e3, 1)})
+               "dpfs_per_sec": round(nb / ms * 1e3, 1)})
 
 
 if __name__ == "__main__":

@@ -28,6 +28,15 @@ void launch_bfs(std::uintptr_t keys, std::uintptr_t out,
                 std::uintptr_t aes_tabs, int batch, long long n, int depth,
                 int prf, std::uintptr_t stream);
 
+// Grid-wide cooperative single-key strategy (the reference's dpf_coop.cu):
+// one cooperative launch walks one key's whole tree with a grid sync per
+// level.  fused=true MACs the leaves against the permuted table into
+// out[16] (zeroed by caller); fused=false writes natural-order one-hot
+// low-32 shares to out[n].
+void launch_coop(std::uintptr_t keys, std::uintptr_t table, std::uintptr_t out,
+                 std::uintptr_t aes_tabs, long long n, int depth, int zlog,
+                 int prf, bool fused, std::uintptr_t stream);
+
 // Naive per-leaf oracle (O(n log n) PRFs), natural order output.  Test-only.
 void launch_naive(std::uintptr_t keys, std::uintptr_t out,
                   std::uintptr_t aes_tabs, int batch, long long n, int depth,

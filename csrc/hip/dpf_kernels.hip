@@ -967,6 +967,109 @@ __global__ __launch_bounds__(256) void dpf_bfs_level_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// Grid-wide cooperative single-key strategy (the reference's dpf_coop.cu:
+// cudaLaunchCooperativeKernel + this_grid().sync() between tree levels,
+// batch=1 only).  The whole grid walks one GGM tree breadth-first with a
+// grid sync per level; with FUSED the leaves are MAC'd against the
+// (leaf_perm-ordered) table into per-thread accumulators and combined
+// with wrapping atomics — no second kernel.
+//
+// This exists for strategy-matrix completeness and as a measured
+// comparison point: the production fused kernel's j-split serves the
+// same single-key-latency role with NO grid-wide synchronization (its
+// segments are independent), which is why it wins (see
+// benchmarks/strategy_compare.py --coop).
+// ---------------------------------------------------------------------------
+#include <hip/hip_cooperative_groups.h>
+
+template <int PRF, bool FUSED>
+__global__ __launch_bounds__(256) void dpf_coop_kernel(
+    const int* __restrict__ keys, const u32* __restrict__ table,
+    u32* __restrict__ out, const u32* __restrict__ aes_tabs,
+    uint4* __restrict__ ping, uint4* __restrict__ pong, int depth, int zlog,
+    long long n) {
+  extern __shared__ u32 smem[];
+  u32* aes_lds = smem;
+  const int t = (int)threadIdx.x;
+  if constexpr (PRF == PRF_AES128) {
+    for (int e = t; e < 256; e += blockDim.x) {
+      const u32 v = aes_tabs[e];
+#pragma unroll
+      for (int c = 0; c < AES_REP; ++c) aes_lds[e * AES_REP + c] = v;
+    }
+    __syncthreads();
+  }
+  AesLds T{aes_lds, (u32)(t & (AES_REP - 1))};
+  auto grid = cooperative_groups::this_grid();
+  const long long gsize = (long long)gridDim.x * blockDim.x;
+  const long long gtid = (long long)blockIdx.x * blockDim.x + t;
+  const uint4* cw = reinterpret_cast<const uint4*>(keys + 4);
+
+  u32 acc[16];
+#pragma unroll
+  for (int w = 0; w < 16; ++w) acc[w] = 0;
+
+  for (int level = 0; level < depth; ++level) {
+    const long long n_parents = (long long)1 << level;
+    const int i_eval = depth - 1 - level;
+    for (long long p = gtid; p < n_parents; p += gsize) {
+      uint4 seed;
+      if (level == 0) {
+        const int* rp = keys + 516;
+        seed = make_uint4((u32)rp[0], (u32)rp[1], (u32)rp[2], (u32)rp[3]);
+      } else {
+        seed = ping[p];
+      }
+      const int sel = (int)(seed.x & 1u);
+      uint4 c0, c1;
+      prf_pair<PRF>(seed, T, c0, c1);
+      c0 = add128(c0, cw[sel * 64 + i_eval * 2 + 0]);
+      c1 = add128(c1, cw[sel * 64 + i_eval * 2 + 1]);
+      if (level < depth - 1) {
+        pong[p] = c0;
+        pong[p + n_parents] = c1;
+      } else {
+        // natural leaf indices: p and p | n/2 (bits consumed LSB-first)
+        const long long i0 = p, i1 = p | (n >> 1);
+        if constexpr (FUSED) {
+#pragma unroll
+          for (int q = 0; q < 2; ++q) {
+            const long long idx = q ? i1 : i0;
+            const u32 v = q ? c1.x : c0.x;
+            // leaf_perm(idx) in-kernel (see dpf_core.cc layout contract)
+            const int ds = depth - zlog;
+            const u32 tl = (u32)(idx & ((1u << zlog) - 1));
+            const u32 tpos = __brev(tl) >> (32 - zlog);
+            const u32 jm = (u32)((idx >> zlog) & (((long long)1 << (ds - 1)) - 1));
+            const u32 j = ds > 1 ? (__brev(jm) >> (33 - ds)) : 0;
+            const u32 b = (u32)(idx >> (depth - 1));
+            const long long row = ((long long)j << (zlog + 1)) |
+                                  ((long long)tpos << 1) | b;
+            const u32* trow = table + (u64)row * 16;
+#pragma unroll
+            for (int w = 0; w < 16; ++w) acc[w] += v * trow[w];
+          }
+        } else {
+          out[i0] = c0.x;
+          out[i1] = c1.x;
+        }
+      }
+    }
+    if (level < depth - 1) {
+      grid.sync();
+      uint4* tmp = ping;
+      ping = pong;
+      pong = tmp;
+    }
+  }
+  if constexpr (FUSED) {
+#pragma unroll
+    for (int w = 0; w < 16; ++w)
+      if (acc[w]) atomicAdd(out + w, acc[w]);
+  }
+}
+
+// ---------------------------------------------------------------------------
 // ALU / PRF probe kernels — the unit-test analog of the reference's
 // dpf_gpu/tests/test_128_bit.cu:192-200 (device add/mul/pack asserted
 // against host __int128): one element per thread, results compared
@@ -1081,6 +1184,72 @@ void launch_probe_prf(std::uintptr_t seeds, std::uintptr_t aes_tabs,
 namespace {
 uint4* get_scratch(size_t bytes, hipStream_t stream);  // defined below
 
+template <int PRF, bool FUSED>
+void launch_coop_t(const int* keys, const u32* table, u32* out,
+                   const u32* aes_tabs, uint4* ping, uint4* pong, int depth,
+                   int zlog, long long n, hipStream_t st) {
+  auto kern = dpf_coop_kernel<PRF, FUSED>;
+  const size_t shmem = (PRF == PRF_AES128) ? AES_LDS_WORDS * 4 : 0;
+  int dev = 0;
+  HIP_CHECK(hipGetDevice(&dev));
+  int coop = 0;
+  HIP_CHECK(hipDeviceGetAttribute(&coop, hipDeviceAttributeCooperativeLaunch,
+                                  dev));
+  if (!coop)
+    throw std::runtime_error("device does not support cooperative launch");
+  int num_cu = 0;
+  HIP_CHECK(hipDeviceGetAttribute(&num_cu,
+                                  hipDeviceAttributeMultiprocessorCount, dev));
+  int max_blocks = 0;
+  HIP_CHECK(hipOccupancyMaxActiveBlocksPerMultiprocessor(
+      &max_blocks, (const void*)kern, 256, shmem));
+  long long grid = (long long)num_cu * max_blocks;
+  const long long need = ((n / 2) + 255) / 256;  // widest level
+  if (grid > need) grid = need;
+  if (grid < 1) grid = 1;
+  void* args[] = {(void*)&keys, (void*)&table, (void*)&out,
+                  (void*)&aes_tabs, (void*)&ping, (void*)&pong,
+                  (void*)&depth, (void*)&zlog, (void*)&n};
+  HIP_CHECK(hipLaunchCooperativeKernel((const void*)kern,
+                                       dim3((unsigned)grid), dim3(256), args,
+                                       shmem, st));
+}
+}  // namespace
+
+void launch_coop(std::uintptr_t keys, std::uintptr_t table, std::uintptr_t out,
+                 std::uintptr_t aes_tabs, long long n, int depth, int zlog,
+                 int prf, bool fused, std::uintptr_t stream) {
+  if (depth < 1 || ((long long)1 << depth) != n)
+    throw std::invalid_argument("bad depth/n");
+  auto st = reinterpret_cast<hipStream_t>(stream);
+  const size_t half = (size_t)(n / 2) * sizeof(uint4);
+  uint4* ping = get_scratch(2 * half > 0 ? 2 * half : 32, st);
+  uint4* pong = ping + (size_t)(n / 2);
+  auto* kp = reinterpret_cast<const int*>(keys);
+  auto* tp = reinterpret_cast<const u32*>(table);
+  auto* op = reinterpret_cast<u32*>(out);
+  auto* ap = reinterpret_cast<const u32*>(aes_tabs);
+#define COOP_CASE(P)                                                       \
+  case P:                                                                  \
+    if (fused)                                                             \
+      launch_coop_t<P, true>(kp, tp, op, ap, ping, pong, depth, zlog, n,  \
+                             st);                                          \
+    else                                                                   \
+      launch_coop_t<P, false>(kp, tp, op, ap, ping, pong, depth, zlog, n, \
+                              st);                                         \
+    break;
+  switch (prf) {
+    COOP_CASE(PRF_DUMMY)
+    COOP_CASE(PRF_SALSA20)
+    COOP_CASE(PRF_CHACHA20)
+    COOP_CASE(PRF_AES128)
+    default:
+      throw std::invalid_argument("unknown PRF");
+  }
+#undef COOP_CASE
+}
+
+namespace {
 template <int PRF>
 void launch_bfs_t(const int* keys, u32* out, const u32* aes_tabs,
                   uint4* ping, uint4* pong, int batch, long long n, int depth,

@@ -71,6 +71,10 @@ PYBIND11_MODULE(_hip, m) {
         py::arg("aes_tabs"), py::arg("batch"), py::arg("n"), py::arg("depth"),
         py::arg("prf"), py::arg("stream"),
         py::call_guard<py::gil_scoped_release>());
+  m.def("eval_coop", &gpudpf_hip::launch_coop, py::arg("keys"),
+        py::arg("table"), py::arg("out"), py::arg("aes_tabs"), py::arg("n"),
+        py::arg("depth"), py::arg("zlog"), py::arg("prf"), py::arg("fused"),
+        py::arg("stream"), py::call_guard<py::gil_scoped_release>());
   m.def("eval_naive", &gpudpf_hip::launch_naive, py::arg("keys"),
         py::arg("out"), py::arg("aes_tabs"), py::arg("batch"), py::arg("n"),
         py::arg("depth"), py::arg("prf"), py::arg("stream"),

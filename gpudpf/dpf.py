@@ -279,6 +279,8 @@ class DPF(object):
             strategy = "two_stage"  # wide entries go through the MFMA path
         if strategy == "two_stage" and not one_hot_only:
             return self._eval_gpu_two_stage(kt, out_device=out_device)
+        if strategy == "coop":
+            return self._eval_gpu_coop(kt, n, depth, one_hot_only, out_device)
         batch = kt.shape[0]
         dev = self._table_gpu.device
         stream = torch.cuda.current_stream(dev).cuda_stream
@@ -377,6 +379,32 @@ class DPF(object):
         torch.cuda.current_stream(dev).wait_stream(s_expand)
         out = torch.cat(outs) if len(outs) > 1 else outs[0]
         out = out[:, : self.table_effective_entry_size]
+        return out if out_device else out.cpu()
+
+    def _eval_gpu_coop(self, kt, n, depth, one_hot_only, out_device):
+        """Grid-wide cooperative strategy (the reference's dpf_coop.cu):
+        one cooperative launch per key, grid sync per tree level.  Kept
+        as a measured research strategy — the production j-split serves
+        the single-key-latency role without grid-wide synchronization."""
+        dev = self._table_gpu.device
+        keys_gpu = kt.to(dev, non_blocking=True).contiguous()
+        b = kt.shape[0]
+        stream = torch.cuda.current_stream(dev).cuda_stream
+        if one_hot_only:
+            out = torch.empty((b, n), dtype=torch.int32, device=dev)
+        else:
+            out = torch.zeros((b, self.ENTRY_SIZE), dtype=torch.int32,
+                              device=dev)
+        for i in range(b):
+            _hip.eval_coop(
+                keys_gpu[i].data_ptr(), self._table_gpu.data_ptr(),
+                out[i].data_ptr(), self._aes_ptr, n, depth, self._zlog,
+                self.prf_method, not one_hot_only, stream,
+            )
+        if one_hot_only:
+            out = out[:, : self.table_num_entries]
+        else:
+            out = out[:, : self.table_effective_entry_size]
         return out if out_device else out.cpu()
 
     def eval_gpu_into(self, keys_gpu, out_gpu):

@@ -85,6 +85,25 @@ def test_gpu_bfs_strategy_matches_cpu():
             assert np.array_equal(got[1], want2), (n, prf)
 
 
+def test_gpu_coop_strategy_matches_fused():
+    # grid-wide cooperative kernel (one cooperative launch per key, grid
+    # sync per level): fused output must equal the production path, and
+    # the one-hot output must match the CPU core in natural order
+    n = 1 << 14
+    for prf in PRFS:
+        dpf = DPF(prf=prf)
+        table = torch.randint(-(2**31), 2**31 - 1, (n, 16),
+                              dtype=torch.int64).to(torch.int32)
+        dpf.eval_init(table)
+        k1, k2 = dpf.gen(9999, n)
+        kt = torch.stack([k1, k2])
+        want = dpf.eval_gpu(kt)
+        got = dpf.eval_gpu(kt, strategy="coop")
+        assert torch.equal(got, want), prf
+        oh = dpf.eval_gpu([k1], one_hot_only=True, strategy="coop").numpy()
+        assert np.array_equal(oh[0], _core.expand(k1.numpy(), prf)), prf
+
+
 def test_gpu_naive_kernel_oracle():
     from gpudpf import _hip
 
