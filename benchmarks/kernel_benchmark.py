@@ -71,6 +71,17 @@ def run(strategy, prf_name, n, batch, entry_size, reps, check=False):
             out = torch.empty((b, n), dtype=torch.int32, device=dev)
             _hip.eval_naive(keys_g.data_ptr(), out.data_ptr(), aes_ptr, b, n,
                             depth, prf, stream)
+        elif strategy == "bfs":
+            out = torch.empty((b, n), dtype=torch.int32, device=dev)
+            _hip.eval_bfs(keys_g.data_ptr(), out.data_ptr(), aes_ptr, b, n,
+                          depth, prf, stream)
+        elif strategy == "coop":
+            out = torch.zeros((b, 16), dtype=torch.int32, device=dev)
+            for i in range(b):
+                _hip.eval_coop(keys_g[i].data_ptr(),
+                               dpf._table_gpu.data_ptr(), out[i].data_ptr(),
+                               aes_ptr, n, depth, dpf._zlog, prf, True,
+                               stream)
         else:
             raise ValueError(strategy)
         return out
@@ -89,9 +100,15 @@ def run(strategy, prf_name, n, batch, entry_size, reps, check=False):
             got = launch(1, keys_gpu[:1])[0].cpu()
             perm = torch.from_numpy(_core.leaf_perm_table(n, dpf._zlog))
             assert torch.equal(got[perm], want_shares), "expand check failed"
-        else:  # naive writes natural order directly
+        elif strategy == "coop":
             got = launch(1, keys_gpu[:1])[0].cpu()
-            assert torch.equal(got, want_shares), "naive check failed"
+            padded = torch.nn.functional.pad(table, (0, 16 - entry_size))
+            want = torch.from_numpy(
+                _core.eval_fused_cpu(k1.numpy(), padded.numpy(), prf))
+            assert torch.equal(got, want), "coop check failed"
+        else:  # naive/bfs write natural order directly
+            got = launch(1, keys_gpu[:1])[0].cpu()
+            assert torch.equal(got, want_shares), "%s check failed" % strategy
 
     # clock-ramp warmup
     tw = time.time()
@@ -131,7 +148,7 @@ def run(strategy, prf_name, n, batch, entry_size, reps, check=False):
 if __name__ == "__main__":
     ap = argparse.ArgumentParser()
     ap.add_argument("--strategy", default="fused",
-                    choices=["fused", "expand", "naive"])
+                    choices=["fused", "expand", "naive", "bfs", "coop"])
     ap.add_argument("--prf", default="AES128", choices=list(PRF_IDS))
     ap.add_argument("--n", type=int, default=65536)
     ap.add_argument("--batch", type=int, default=512)

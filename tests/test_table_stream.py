@@ -2,6 +2,8 @@
 table_read, reconstruction through both the fused and two-stage paths,
 and the chunk-wise (non-materialized) permutation used for huge domains."""
 
+import os
+
 import numpy as np
 import pytest
 import torch
@@ -86,3 +88,22 @@ def test_chunkwise_perm_path():
             d2.eval_gpu([k1], one_hot_only=True)
     finally:
         DPF.PERM_MATERIALIZE_MAX = old
+
+
+def test_huge_table_script_smoke():
+    """The huge-table benchmark script's contract (fill -> verify ->
+    serve) at the smoke shape, run exactly as shipped."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "benchmarks/huge_table.py", "--shape", "smoke",
+         "--steps", "2", "--warmup", "1"],
+        cwd=repo, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stdout + r.stderr
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert any("'verified'" in l for l in lines), r.stdout
+    final = eval(lines[-1], {"__builtins__": {}})  # dict-line contract
+    assert final["shape"] == "smoke" and final["ms_per_step"] > 0
