@@ -79,6 +79,39 @@ class ShardedDPF(object):
         self.table_effective_entry_size = int(local_rows.shape[1])
         self.local.eval_init(local_rows.contiguous())
 
+    def eval_init_empty(self, n, e):
+        """Streaming-ingest mode for aggregate tables near W x HBM: each
+        rank allocates its n/W-row shard directly in device memory (never
+        on the host) and fills it with table_write."""
+        n = int(n)
+        if n & (n - 1) != 0:
+            raise Exception("sharded tables must have power-of-two entries")
+        if n % self.world != 0:
+            raise Exception("n must be divisible by world size")
+        self.table_num_entries = n
+        self.table_effective_entry_size = int(e)
+        self.local.eval_init_empty(n // self.world, e)
+
+    def table_write(self, indices, rows):
+        """Write natural GLOBAL-index rows; each rank keeps its residue
+        class (idx %% W == rank) and ignores the rest, so every rank can
+        be fed the same stream."""
+        idx = indices if isinstance(indices, torch.Tensor) \
+            else torch.as_tensor(indices, dtype=torch.int64)
+        mine = (idx % self.world) == self.rank
+        if not bool(mine.any()):
+            return
+        self.local.table_write(idx[mine] // self.world, rows[mine])
+
+    def table_read(self, indices):
+        """Gather natural global rows owned by THIS rank (indices must
+        all satisfy idx %% W == rank)."""
+        idx = indices if isinstance(indices, torch.Tensor) \
+            else torch.as_tensor(indices, dtype=torch.int64)
+        if bool(((idx % self.world) != self.rank).any()):
+            raise Exception("table_read: some indices belong to other ranks")
+        return self.local.table_read(idx // self.world)
+
     def shard_subkeys(self, keys):
         """Restrict a batch of full-domain wire-format keys to this rank's
         residue class: [b, 524] int32 CPU tensor of depth-log2(W) subkeys.

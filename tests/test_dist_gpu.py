@@ -176,3 +176,51 @@ def test_sharded_wide_entries_rccl():
     rank, ok, err = q.get(timeout=300)
     p.join(timeout=60)
     assert ok, f"rank {rank}: {err}"
+
+
+def _nccl_stream_ingest_worker(port, q):
+    """Sharded streaming ingest (eval_init_empty + global-index
+    table_write): shard rows never exist on the host as a full table."""
+    import torch.distributed as td
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    td.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        from gpudpf import DPF, ShardedDPF, _core
+
+        N, e = 1 << 13, 16
+        prf = DPF.PRF_AES128
+        torch.manual_seed(6)
+        table = torch.randint(-(2**31), 2**31 - 1, (N, e),
+                              dtype=torch.int64).to(torch.int32)
+        sd = ShardedDPF(prf=prf, device="cuda:0")
+        sd.eval_init_empty(N, e)
+        for lo in range(0, N, 1000):  # global-index streaming fill
+            hi = min(N, lo + 1000)
+            sd.table_write(torch.arange(lo, hi), table[lo:hi])
+        idxs = [2, 777, N - 1]
+        back = sd.table_read(torch.tensor(
+            [i for i in idxs if i % sd.world == sd.rank]))
+        ok = back.shape[1] == e
+        ks = [(_core.gen(i, N, b"si-%d" % i, prf)) for i in idxs]
+        a = sd.eval_gpu([torch.from_numpy(k[0]) for k in ks])
+        b = sd.eval_gpu([torch.from_numpy(k[1]) for k in ks])
+        rec = (a.to(torch.int64) - b.to(torch.int64)).to(
+            torch.int32).numpy()
+        ok = ok and bool(np.array_equal(rec, table[idxs, :].numpy()))
+        q.put((0, ok, ""))
+    except Exception as exc:  # pragma: no cover
+        q.put((0, False, repr(exc)))
+    finally:
+        td.destroy_process_group()
+
+
+def test_sharded_streaming_ingest_rccl():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    p = ctx.Process(target=_nccl_stream_ingest_worker, args=(29801, q))
+    p.start()
+    rank, ok, err = q.get(timeout=300)
+    p.join(timeout=60)
+    assert ok, f"rank {rank}: {err}"
