@@ -146,8 +146,10 @@ class ShardedDPF(object):
             part = self._allreduce_(part)
             part = part[:, : self.table_effective_entry_size]
         else:
-            # wide entries: two-stage MFMA path (device-resident output)
-            part = self.local.eval_gpu(subs, out_device=True)
+            # wide entries: two-stage GEMM path (device-resident output);
+            # the trimmed [b, e] view of the padded output is not
+            # contiguous, and RCCL collectives need dense buffers
+            part = self.local.eval_gpu(subs, out_device=True).contiguous()
             part = self._allreduce_(part)
         return part.cpu() if to_host else part
 
