@@ -64,7 +64,7 @@ namespace {
 
 constexpr int kThreads = 64;       // one wave per workgroup
 constexpr int kColsPerTile = 256;  // 64 lanes x 4 columns
-constexpr int kChunkK = 32;        // staged share rows per LDS refill
+constexpr int kChunkK = 64;        // staged share rows per LDS refill
 constexpr int kMaxB = 16;          // accumulators = 4*B VGPRs per lane
 
 // Grow-only per-device partials scratch (kept alive forever so hipGraphs
@@ -162,7 +162,7 @@ __global__ __launch_bounds__(kThreads) void gemm_u32_stream_kernel(
     if (in_n) {
       const u32* brow = b + (u64)kc * (u64)N + (u64)col;
       if (kn == kChunkK) {
-#pragma unroll 4
+#pragma unroll 8
         for (int kk = 0; kk < kChunkK; ++kk) {
           const uint4 v = *reinterpret_cast<const uint4*>(brow + (u64)kk * N);
 #pragma unroll
