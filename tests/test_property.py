@@ -91,3 +91,56 @@ def test_aes_is_permutation_like(data, key):
     assert c1 == c2
     flipped = bytes([key[0] ^ 1]) + key[1:]
     assert _core.aes_block(flipped, data) != c1
+
+
+@settings(max_examples=25, deadline=None)
+@given(
+    depth=st.integers(min_value=7, max_value=20),
+    alpha_frac=st.floats(min_value=0.0, max_value=0.999),
+    prf=st.sampled_from(PRFS),
+)
+def test_compact_key_round_trip_property(depth, alpha_frac, prf):
+    n = 1 << depth
+    alpha = int(alpha_frac * n)
+    k1, k2 = _core.gen(alpha, n, b"cp-%d-%d" % (depth, alpha), prf)
+    for k in (k1, k2):
+        c = _core.key_compact(k)
+        assert c.nbytes == (3 + 4 * depth) * 16
+        back = _core.key_expand_compact(c)
+        assert np.array_equal(back, k)
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    logw=st.integers(min_value=0, max_value=4),
+    alpha_frac=st.floats(min_value=0.0, max_value=0.999),
+    prf=st.sampled_from(PRFS),
+)
+def test_shard_subkey_batch_property(logw, alpha_frac, prf):
+    n, world = 1 << 12, 1 << logw
+    alpha = int(alpha_frac * n)
+    k1, _ = _core.gen(alpha, n, b"sb-%d" % alpha, prf)
+    batch = np.stack([k1, k1])
+    for rank in range(world):
+        got = _core.shard_subkey_batch(batch, prf, rank, world)
+        want = _core.shard_subkey(k1, prf, rank, world)
+        assert np.array_equal(got[0], want) and np.array_equal(got[1], want)
+        # the subkey really evaluates the residue class
+        sub_oh = _core.expand(got[0], prf)
+        full_oh = _core.expand(k1, prf)
+        assert np.array_equal(sub_oh, full_oh[rank::world])
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    logn=st.integers(min_value=7, max_value=18),
+    frac=st.floats(min_value=0.0, max_value=0.999),
+)
+def test_leaf_perm_rows_matches_table(logn, frac):
+    n = 1 << logn
+    zlog = _core.zlog_for_depth(logn)
+    idxs = np.unique(np.array(
+        [0, n - 1, int(frac * n), (int(frac * n) * 7919) % n], dtype=np.int64))
+    rows = _core.leaf_perm_rows(idxs, n, zlog)
+    full = _core.leaf_perm_table(n, zlog)
+    assert np.array_equal(rows, full[idxs])
