@@ -17,6 +17,7 @@ Public API parity (reference dpf.py:35-137):
 try:
     from gpudpf.dpf import DPF  # noqa: F401
     from gpudpf.dist import ReplicatedDPF, ShardedDPF  # noqa: F401
+    from gpudpf.serving import GraphedServer  # noqa: F401
 except ImportError:
     # Fresh checkout: the native extensions are not built yet.  Importing
     # the bare package must still work so `gpudpf._build` can bootstrap
