@@ -85,6 +85,37 @@ def test_gpu_bfs_strategy_matches_cpu():
             assert np.array_equal(got[1], want2), (n, prf)
 
 
+def test_gpu_pipelined_server_interleaved_batches():
+    # double-buffered serving: interleaved submits of DIFFERENT batches
+    # must return each batch's own correct shares
+    from gpudpf.serving import PipelinedServer
+
+    n, batch = 16384, 64
+    dpf = DPF(prf=DPF.PRF_SALSA20)
+    table = torch.randint(-(2**31), 2**31 - 1, (n, 16),
+                          dtype=torch.int64).to(torch.int32)
+    dpf.eval_init(table)
+    srv = PipelinedServer(dpf, batch, depth=2)
+    batches = []
+    for r in range(6):
+        ks = torch.stack([dpf.gen((r * 1000 + i * 13) % n, n)[0]
+                          for i in range(batch)])
+        batches.append(ks)
+    pending = []
+    results = []
+    for ks in batches:
+        pending.append((srv.submit(ks), ks))
+        if len(pending) >= 2:
+            h, k = pending.pop(0)
+            results.append((srv.collect(h), k))
+    while pending:
+        h, k = pending.pop(0)
+        results.append((srv.collect(h), k))
+    for got, ks in results:
+        want = dpf.eval_gpu(ks)
+        assert torch.equal(got, want)
+
+
 def test_gpu_coop_strategy_matches_fused():
     # grid-wide cooperative kernel (one cooperative launch per key, grid
     # sync per level): fused output must equal the production path, and
