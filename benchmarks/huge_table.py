@@ -56,6 +56,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--prf", default="CHACHA20")
     ap.add_argument("--checks", type=int, default=4)
+    ap.add_argument("--pipeline", action="store_true",
+                    help="pipelined wide serving (TwoStageServer)")
     args = ap.parse_args()
 
     log_n, e = SHAPES[args.shape]
@@ -101,12 +103,37 @@ def main():
         torch.randint(0, n, (args.batch,), generator=g).numpy(),
         n, b"huge-bench", prf)
     keys = torch.from_numpy(ks)
+    ep = -(-e // 16) * 16
+    if args.pipeline and ep > 16 and args.batch <= 16:
+        # pipelined wide serving: expansion of step i+1 under the GEMM
+        # of step i (every step still does its full work)
+        from gpudpf.serving import TwoStageServer
+
+        srv = TwoStageServer(d, args.batch)
+        pending = []
+
+        def step():
+            pending.append(srv.submit(keys))
+            if len(pending) >= 2:
+                srv.collect(pending.pop(0))
+
+        def drain():
+            while pending:
+                srv.collect(pending.pop(0))
+    else:
+        def step():
+            d.eval_gpu(keys)
+
+        def drain():
+            pass
     for _ in range(args.warmup):
-        d.eval_gpu(keys)
+        step()
+    drain()
     torch.cuda.synchronize()
     t0 = time.time()
     for _ in range(args.steps):
-        d.eval_gpu(keys)
+        step()
+    drain()
     torch.cuda.synchronize()
     dt = (time.time() - t0) / args.steps
     # effective table traffic of the timed path: the two-stage GEMM

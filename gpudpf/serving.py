@@ -227,3 +227,100 @@ class PipelinedServer:
         slot["busy"] = False
         return slot["out_pinned"][:, : self.dpf.table_effective_entry_size
                                   ].clone()
+
+
+class TwoStageServer:
+    """Pipelined wide-entry serving (entry > 16 words): batch i+1's
+    one-hot expansion runs on its own stream while batch i's streaming
+    GEMM reads the table — steady-state step = max(expansion, GEMM)
+    instead of their sum.  On a 240 GB table (2^28 x 224) the serial
+    two-stage step is ~53 ms expansion + ~64 ms GEMM; pipelined serving
+    approaches the larger of the two.
+
+    Usage:
+        srv = TwoStageServer(dpf, batch=8)   # dpf eval_init'ed, wide
+        h = srv.submit(keys_cpu)             # non-blocking
+        shares = srv.collect(h)              # [batch, e] int32 CPU
+    """
+
+    def __init__(self, dpf: DPF, batch: int, depth: int = 2):
+        if dpf._table_gpu is None:
+            raise Exception("eval_init the DPF before building a server")
+        if dpf._entry_padded <= DPF.ENTRY_SIZE:
+            raise Exception("TwoStageServer serves wide entries "
+                            "(> 16 words); use PipelinedServer")
+        if batch > 16:
+            raise Exception("streaming-GEMM batch is capped at 16 per "
+                            "table pass; lower the batch")
+        self.dpf = dpf
+        self.batch = batch
+        self.depth = depth
+        dev = dpf._table_gpu.device
+        self.device = dev
+        n = dpf._n_domain
+        ep = dpf._entry_padded
+        free, _total = torch.cuda.mem_get_info(dev)
+        need = depth * batch * n * 4
+        if need > free - (2 << 30):
+            raise Exception("pipeline share buffers need %d bytes "
+                            "(free %d); lower batch/depth" % (need, free))
+        self._gemm_stream = torch.cuda.Stream(dev)
+        self._slots = []
+        for _ in range(depth):
+            self._slots.append({
+                "stream": torch.cuda.Stream(dev),
+                "keys_pinned": torch.zeros((batch, DPF.KEY_INTS),
+                                           dtype=torch.int32).pin_memory(),
+                "keys_gpu": torch.zeros((batch, DPF.KEY_INTS),
+                                        dtype=torch.int32, device=dev),
+                "shares": torch.zeros((batch, n), dtype=torch.int32,
+                                      device=dev),
+                "out_gpu": torch.zeros((batch, ep), dtype=torch.int32,
+                                       device=dev),
+                "ev_expand": torch.cuda.Event(),
+                "ev_done": torch.cuda.Event(),
+                "busy": False,
+            })
+        self._next = 0
+
+    def submit(self, keys):
+        if not isinstance(keys, torch.Tensor):
+            keys = torch.stack([k.reshape(-1) for k in keys])
+        if keys.shape[0] != self.batch:
+            raise Exception("TwoStageServer is fixed at batch=%d"
+                            % self.batch)
+        dpf = self.dpf
+        slot = self._slots[self._next]
+        self._next = (self._next + 1) % self.depth
+        if slot["busy"]:
+            slot["ev_done"].synchronize()
+            slot["busy"] = False
+        slot["keys_pinned"].numpy()[:] = keys.numpy()
+        n = dpf._n_domain
+        with torch.cuda.stream(slot["stream"]):
+            slot["keys_gpu"].copy_(slot["keys_pinned"], non_blocking=True)
+            _hip.eval_expand(slot["keys_gpu"].data_ptr(),
+                             slot["shares"].data_ptr(), dpf._aes_ptr,
+                             self.batch, n, dpf._depth, dpf._zlog,
+                             dpf.prf_method, slot["stream"].cuda_stream)
+            slot["ev_expand"].record(slot["stream"])
+        # the GEMM stream serializes table passes across slots (the table
+        # stream is the contended resource); it only waits for THIS
+        # slot's expansion
+        self._gemm_stream.wait_event(slot["ev_expand"])
+        with torch.cuda.stream(self._gemm_stream):
+            slot["out_gpu"].zero_()
+            _hip.gemm_u32_stream(slot["shares"].data_ptr(),
+                                 dpf._table_gpu.data_ptr(),
+                                 slot["out_gpu"].data_ptr(), self.batch, n,
+                                 dpf._entry_padded,
+                                 self._gemm_stream.cuda_stream)
+            slot["ev_done"].record(self._gemm_stream)
+        slot["busy"] = True
+        return slot
+
+    def collect(self, slot):
+        slot["ev_done"].synchronize()
+        slot["busy"] = False
+        return slot["out_gpu"][:, : self.dpf.table_effective_entry_size
+                               ].cpu()

@@ -116,6 +116,32 @@ def test_gpu_pipelined_server_interleaved_batches():
         assert torch.equal(got, want)
 
 
+def test_gpu_two_stage_server_pipelined():
+    # wide-entry pipelined serving: expansion of batch i+1 overlaps the
+    # GEMM of batch i; results must match the plain two-stage path
+    from gpudpf.serving import TwoStageServer
+
+    n, e, batch = 8192, 48, 8
+    dpf = DPF(prf=DPF.PRF_CHACHA20)
+    table = torch.randint(-(2**31), 2**31 - 1, (n, e),
+                          dtype=torch.int64).to(torch.int32)
+    dpf.eval_init(table)
+    srv = TwoStageServer(dpf, batch)
+    batches = [torch.stack([dpf.gen((r * 311 + i * 7) % n, n)[0]
+                            for i in range(batch)]) for r in range(5)]
+    pending, results = [], []
+    for ks in batches:
+        pending.append((srv.submit(ks), ks))
+        if len(pending) >= 2:
+            h, k = pending.pop(0)
+            results.append((srv.collect(h), k))
+    while pending:
+        h, k = pending.pop(0)
+        results.append((srv.collect(h), k))
+    for got, ks in results:
+        assert torch.equal(got, dpf.eval_gpu(ks))
+
+
 def test_gpu_coop_strategy_matches_fused():
     # grid-wide cooperative kernel (one cooperative launch per key, grid
     # sync per level): fused output must equal the production path, and
