@@ -78,3 +78,32 @@ def test_pir_http_roundtrip_gpu():
     b = post([p[1] for p in pairs])
     rec = (a.to(torch.int64) - b.to(torch.int64)).to(torch.int32)
     assert torch.equal(rec, table[idxs, :])
+
+
+def test_pir_fetch_client_helper(monkeypatch):
+    """The 2-server client helper end to end: pir_fetch's HTTP calls are
+    routed to two in-process TestClient apps (distinct trust domains)."""
+    import gpudpf.server as srv_mod
+
+    n, e = 1024, 8
+    torch.manual_seed(3)
+    table = torch.randint(-(2**31), 2**31 - 1, (n, e), dtype=torch.int64).to(
+        torch.int32)
+    apps = {
+        "http://server-a": TestClient(build_app(table=table,
+                                                prf=DPF.PRF_SALSA20)),
+        "http://server-b": TestClient(build_app(table=table,
+                                                prf=DPF.PRF_SALSA20)),
+    }
+
+    class _FakeHttpx:
+        @staticmethod
+        def post(url, json=None, timeout=None):
+            base, path = url.rsplit("/", 1)
+            return apps[base].post("/" + path, json=json)
+
+    monkeypatch.setitem(__import__("sys").modules, "httpx", _FakeHttpx)
+    idxs = [0, 511, 1023]
+    got = srv_mod.pir_fetch(["http://server-a", "http://server-b"], n, idxs,
+                            prf=DPF.PRF_SALSA20)
+    assert torch.equal(got, table[idxs, :])
