@@ -133,7 +133,9 @@ def main():
         engine = DPF(prf=prf, device=device)
         engine.eval_init(table)
         if on_gpu:
-            if args.no_graph:
+            if args.no_graph or args.entry_size > 16:
+                # wide entries route through the two-stage path; plain
+                # per-step eval (the pipeline classes serve fixed shapes)
                 def step(k=keys_cpu):
                     return engine.eval_gpu(k)
                 def drain():
