@@ -66,3 +66,16 @@ def test_bench_world2_gloo_check(mode):
     else:
         assert rec["scaling"] == "weak"
         assert rec["config"]["global_batch"] == 16
+
+
+def test_bench_single_process_cpu_wide_entries():
+    # wide entry sizes must not crash the serving-mode selection
+    cmd = [sys.executable, "bench.py", "--device", "cpu",
+           "--entries", "4096", "--batch", "4", "--steps", "1",
+           "--warmup", "0", "--entry-size", "24", "--prf", "SALSA20",
+           "--check"]
+    r = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    rec = _last_json_line(r.stdout)
+    assert rec["config"]["entry_size"] == 24
