@@ -212,7 +212,8 @@ def main():
     # mode the whole world serves one batch cooperatively.
     batches_per_step = world if args.mode == "replicate" else 1
     value = args.batch * batches_per_step * args.steps / elapsed
-    base = V100_BASELINE.get((args.prf, n))
+    # published reference rows are entry_size=16 only
+    base = V100_BASELINE.get((args.prf, n)) if args.entry_size == 16 else None
 
     if rank == 0:
         print(json.dumps({
