@@ -147,7 +147,9 @@ def main():
                 # interleave, dpf_benchmark.cu:191-231).  Every batch
                 # still does the full step: key H2D, kernel, share D2H.
                 from gpudpf.serving import PipelinedServer
-                srv = PipelinedServer(engine, args.batch, depth=2)
+                srv = PipelinedServer(
+                    engine, args.batch,
+                    depth=int(os.environ.get("GPUDPF_PIPE_DEPTH", "2")))
                 pending = []
                 def step(k=keys_cpu):
                     pending.append(srv.submit(k))
