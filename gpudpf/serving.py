@@ -106,7 +106,10 @@ class GraphedServer:
         torch.cuda.current_stream(dev).wait_stream(s)
         torch.cuda.synchronize(dev)
         self._graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self._graph):
+        # thread_local: the default 'global' capture mode errors ANY other
+        # thread's CUDA call during capture — under a nccl process group
+        # the RCCL watchdog thread races that window
+        with torch.cuda.graph(self._graph, capture_error_mode="thread_local"):
             launch()
 
     def eval(self, keys, to_host=True):
@@ -194,7 +197,8 @@ class PipelinedServer:
             torch.cuda.current_stream(dev).wait_stream(warm)
             torch.cuda.synchronize(dev)
             slot["graph"] = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(slot["graph"], stream=slot["stream"]):
+            with torch.cuda.graph(slot["graph"], stream=slot["stream"],
+                                  capture_error_mode="thread_local"):
                 launch()
             self._slots.append(slot)
         self._next = 0
