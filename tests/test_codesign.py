@@ -54,3 +54,21 @@ def test_join_pareto_marks_lower_ppl_and_higher_throughput():
             if p["config"]["num_bins"] == 16
             and p["config"]["queries_per_bin"] == 1]
     assert best and best[0]["pareto"]
+
+
+def test_plot_accuracy_and_codesign_smoke(tmp_path):
+    """Plotters render from committed sweep/join artifacts."""
+    import os
+
+    from pir import plots
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sweep_dir = os.path.join(repo, "profiles", "codesign", "lm_sweep")
+    join_json = os.path.join(repo, "profiles", "codesign",
+                             "codesign_lm_r2.json")
+    out1 = str(tmp_path / "acc.png")
+    out2 = str(tmp_path / "cd.png")
+    plots.plot_accuracy(sweep_dir, out1)
+    plots.plot_codesign(join_json, out2)
+    assert os.path.getsize(out1) > 1000
+    assert os.path.getsize(out2) > 1000
