@@ -136,12 +136,10 @@ def main():
     drain()
     torch.cuda.synchronize()
     dt = (time.time() - t0) / args.steps
-    # effective table traffic of the timed path: the two-stage GEMM
-    # streams the padded table once per 64-key chunk; the fused path
-    # streams it once per key (with cross-key L2 reuse on top)
-    ep = -(-e // 16) * 16
     # unambiguous lower bound on the table streaming rate: every step
-    # reads the full padded table at least once
+    # reads the full padded table at least once (the two-stage GEMM
+    # streams it once per 16-key chunk; the fused path once per key with
+    # cross-key L2 reuse on top)
     one_pass_gbps = (1 << log_n) * ep * 4 / dt / 1e9
     print({"shape": args.shape, "n": n, "entry_words": e,
            "table_gb": round(table_gb, 1), "prf": args.prf,
