@@ -45,7 +45,7 @@ def _worker(rank, world, port, q):
         td.destroy_process_group()
 
 
-@pytest.mark.parametrize("world", [2, 4])
+@pytest.mark.parametrize("world", [2, 4, 8])
 def test_sharded_eval_matches_single(world):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
